@@ -1,0 +1,212 @@
+/* Fake libamdhip64: hardware-free backend for interceptor tests.
+ *
+ * Pattern borrowed from the reference's test strategy (SURVEY.md §2.4/§4:
+ * the cndev C mock driven by env fixtures): a JSON-free fake HIP runtime
+ * configured by env vars that the CPU-only CI links/preloads in place of
+ * the real library.  Built with SONAME libamdhip64.so so the same test
+ * binaries resolve the real runtime on a GPU box and this fake under
+ * LD_LIBRARY_PATH here.
+ *
+ * Env:
+ *   FAKE_HIP_DEVICES     device count (default 1)
+ *   FAKE_HIP_TOTAL_MEM   bytes per device (default 288 GiB, the MI355X HBM3E)
+ */
+#define _GNU_SOURCE
+#include <pthread.h>
+#include <stdint.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+
+typedef int hipError_t;
+#define hipSuccess 0
+#define hipErrorInvalidValue 1
+#define hipErrorOutOfMemory 2
+
+#define MAX_DEV 16
+static __thread int t_current = 0;
+static uint64_t g_alloc[MAX_DEV];
+static uint64_t g_launches;
+static uint64_t g_managed_allocs;
+static pthread_mutex_t g_mu = PTHREAD_MUTEX_INITIALIZER;
+
+static int dev_count(void) {
+  const char *e = getenv("FAKE_HIP_DEVICES");
+  int n = e ? atoi(e) : 1;
+  return n > 0 && n <= MAX_DEV ? n : 1;
+}
+
+static uint64_t total_mem(void) {
+  const char *e = getenv("FAKE_HIP_TOTAL_MEM");
+  return e ? strtoull(e, NULL, 10) : (288ULL << 30);
+}
+
+/* header: 16 bytes before the returned pointer record {size, dev} */
+typedef struct { uint64_t size; uint64_t dev; } hdr_t;
+
+hipError_t hipGetDeviceCount(int *count) {
+  if (!count) return hipErrorInvalidValue;
+  *count = dev_count();
+  return hipSuccess;
+}
+
+hipError_t hipGetDevice(int *dev) {
+  if (!dev) return hipErrorInvalidValue;
+  *dev = t_current;
+  return hipSuccess;
+}
+
+hipError_t hipSetDevice(int dev) {
+  if (dev < 0 || dev >= dev_count()) return hipErrorInvalidValue;
+  t_current = dev;
+  return hipSuccess;
+}
+
+static hipError_t fake_alloc(void **ptr, size_t size, int managed) {
+  if (!ptr) return hipErrorInvalidValue;
+  pthread_mutex_lock(&g_mu);
+  if (g_alloc[t_current] + size > total_mem()) {
+    pthread_mutex_unlock(&g_mu);
+    return hipErrorOutOfMemory;
+  }
+  hdr_t *h = malloc(sizeof(hdr_t) + (size < (1 << 20) ? size : 0) + 16);
+  if (!h) {
+    pthread_mutex_unlock(&g_mu);
+    return hipErrorOutOfMemory;
+  }
+  h->size = size;
+  h->dev = t_current;
+  g_alloc[t_current] += size;
+  if (managed) g_managed_allocs++;
+  pthread_mutex_unlock(&g_mu);
+  *ptr = (void *)(h + 1);
+  return hipSuccess;
+}
+
+hipError_t hipMalloc(void **ptr, size_t size) { return fake_alloc(ptr, size, 0); }
+
+hipError_t hipExtMallocWithFlags(void **ptr, size_t size, unsigned flags) {
+  (void)flags;
+  return fake_alloc(ptr, size, 0);
+}
+
+hipError_t hipMallocManaged(void **ptr, size_t size, unsigned flags) {
+  (void)flags;
+  return fake_alloc(ptr, size, 1);
+}
+
+hipError_t hipMallocAsync(void **ptr, size_t size, void *stream) {
+  (void)stream;
+  return fake_alloc(ptr, size, 0);
+}
+
+hipError_t hipMallocFromPoolAsync(void **ptr, size_t size, void *pool,
+                                  void *stream) {
+  (void)pool; (void)stream;
+  return fake_alloc(ptr, size, 0);
+}
+
+hipError_t hipMallocPitch(void **ptr, size_t *pitch, size_t width,
+                          size_t height) {
+  if (!pitch) return hipErrorInvalidValue;
+  size_t p = (width + 255) & ~(size_t)255;
+  *pitch = p;
+  return fake_alloc(ptr, p * height, 0);
+}
+
+hipError_t hipFree(void *ptr) {
+  if (!ptr) return hipSuccess;
+  hdr_t *h = ((hdr_t *)ptr) - 1;
+  pthread_mutex_lock(&g_mu);
+  if (g_alloc[h->dev] >= h->size) g_alloc[h->dev] -= h->size;
+  pthread_mutex_unlock(&g_mu);
+  free(h);
+  return hipSuccess;
+}
+
+hipError_t hipFreeAsync(void *ptr, void *stream) {
+  (void)stream;
+  return hipFree(ptr);
+}
+
+hipError_t hipMemGetInfo(size_t *free_out, size_t *total_out) {
+  pthread_mutex_lock(&g_mu);
+  uint64_t used = g_alloc[t_current];
+  pthread_mutex_unlock(&g_mu);
+  uint64_t total = total_mem();
+  if (free_out) *free_out = total > used ? total - used : 0;
+  if (total_out) *total_out = total;
+  return hipSuccess;
+}
+
+hipError_t hipDeviceTotalMem(size_t *bytes, int dev) {
+  (void)dev;
+  if (bytes) *bytes = total_mem();
+  return hipSuccess;
+}
+
+/* matches the R0600 prefix the interceptor patches */
+typedef struct {
+  char name[256];
+  char uuid[16];
+  char luid[8];
+  unsigned luidDeviceNodeMask;
+  size_t totalGlobalMem;
+  char rest[1024];
+} fake_prop_t;
+
+hipError_t hipGetDevicePropertiesR0600(void *prop, int dev) {
+  if (!prop || dev < 0 || dev >= dev_count()) return hipErrorInvalidValue;
+  fake_prop_t *p = (fake_prop_t *)prop;
+  memset(p, 0, sizeof(fake_prop_t));
+  snprintf(p->name, sizeof(p->name), "AMD Instinct MI355X (fake)");
+  p->totalGlobalMem = total_mem();
+  return hipSuccess;
+}
+
+hipError_t hipGetDeviceProperties(void *prop, int dev) {
+  /* legacy struct: name[256] then totalGlobalMem */
+  if (!prop || dev < 0 || dev >= dev_count()) return hipErrorInvalidValue;
+  char *c = (char *)prop;
+  memset(c, 0, 256 + sizeof(size_t));
+  snprintf(c, 256, "AMD Instinct MI355X (fake)");
+  uint64_t t = total_mem();
+  memcpy(c + 256, &t, sizeof(t));
+  return hipSuccess;
+}
+
+typedef struct { unsigned x, y, z; } vdim3;
+
+hipError_t hipLaunchKernel(const void *f, vdim3 grid, vdim3 block, void **args,
+                           size_t shared, void *stream) {
+  (void)f; (void)grid; (void)block; (void)args; (void)shared; (void)stream;
+  __atomic_fetch_add(&g_launches, 1, __ATOMIC_RELAXED);
+  return hipSuccess;
+}
+
+hipError_t hipModuleLaunchKernel(void *f, unsigned gx, unsigned gy, unsigned gz,
+                                 unsigned bx, unsigned by, unsigned bz,
+                                 unsigned shared, void *stream, void **params,
+                                 void **extra) {
+  (void)f; (void)gx; (void)gy; (void)gz; (void)bx; (void)by; (void)bz;
+  (void)shared; (void)stream; (void)params; (void)extra;
+  __atomic_fetch_add(&g_launches, 1, __ATOMIC_RELAXED);
+  return hipSuccess;
+}
+
+hipError_t hipDeviceSynchronize(void) { return hipSuccess; }
+hipError_t hipStreamSynchronize(void *s) { (void)s; return hipSuccess; }
+
+/* test introspection */
+uint64_t fake_hip_launch_count(void) {
+  return __atomic_load_n(&g_launches, __ATOMIC_RELAXED);
+}
+uint64_t fake_hip_managed_count(void) {
+  return __atomic_load_n(&g_managed_allocs, __ATOMIC_RELAXED);
+}
+uint64_t fake_hip_device_usage(int dev) {
+  pthread_mutex_lock(&g_mu);
+  uint64_t u = g_alloc[dev];
+  pthread_mutex_unlock(&g_mu);
+  return u;
+}

@@ -1,0 +1,295 @@
+/* CU-percent soft throttle: token bucket on kernel launches, driven by a
+ * utilization-feedback watcher thread.
+ *
+ * MI355X enforcement is two-layered (SURVEY.md §7 hard part 2):
+ *  - HARD floor: the device plugin sets HSA_CU_MASK at Allocate so ROCr
+ *    creates every queue on a fixed CU subset of the 256-CU chip (the DCU
+ *    vdev cu_mask design, reference pkg/device-plugin/hygon/dcu/corealloc.go)
+ *    — no interceptor work needed, enforced by the runtime.
+ *  - SOFT ceiling (this file): a shared token bucket in the region paced to
+ *    the container's CU percent.  Launch credits are workgroup counts; a
+ *    watcher thread measures GPU utilization and adjusts the fill rate so
+ *    the container converges on its share even when the CU mask is disabled
+ *    or work is launch-bound.  Reference analog: rate_limiter +
+ *    utilization_watcher in the CUDA hook (SURVEY.md §2.6 "Core throttle").
+ *
+ * Tunables (env):
+ *  VGPU_TOKEN_RATE   fixed tokens/sec, disables feedback (deterministic tests)
+ *  VGPU_UTIL_FILE    file with "<percent>" to read utilization from (tests)
+ *  VGPU_SYSFS_CARDS  comma list of drm card names by visible device index
+ */
+#define _GNU_SOURCE
+#include "vgpu.h"
+
+#include <dirent.h>
+#include <fcntl.h>
+#include <pthread.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <time.h>
+#include <unistd.h>
+
+typedef int hipError_t;
+#define hipSuccess 0
+#define hipErrorInvalidValue 1
+
+#define NSEC 1000000000ULL
+#define REFILL_INTERVAL_NS (50ULL * 1000 * 1000) /* 50 ms */
+/* A full MI355X retires small workgroups at O(10^6)/s when launch-bound;
+ * credits are deliberately generous — the throttle should bind utilization,
+ * not launch count, and the feedback loop re-scales it anyway. */
+#define RATE_FULL 4000000.0 /* workgroups/sec at 100% */
+#define BUCKET_SECONDS 0.25 /* cap: a quarter second of fill */
+
+static pthread_t g_watcher;
+static int g_watcher_started = 0;
+static pthread_mutex_t g_watch_mu = PTHREAD_MUTEX_INITIALIZER;
+static double g_rate_scale[VGPU_MAX_DEVICES]; /* feedback multiplier */
+static char g_card_path[VGPU_MAX_DEVICES][256];
+static int g_cards_resolved = 0;
+
+static uint64_t now_ns(void) {
+  struct timespec ts;
+  clock_gettime(CLOCK_MONOTONIC, &ts);
+  return (uint64_t)ts.tv_sec * NSEC + ts.tv_nsec;
+}
+
+static void resolve_cards(void) {
+  if (g_cards_resolved) return;
+  g_cards_resolved = 1;
+  const char *env = getenv("VGPU_SYSFS_CARDS");
+  if (env && *env) {
+    char tmp[1024];
+    strncpy(tmp, env, sizeof(tmp) - 1);
+    tmp[sizeof(tmp) - 1] = 0;
+    char *save = NULL;
+    int i = 0;
+    for (char *tok = strtok_r(tmp, ",", &save); tok && i < VGPU_MAX_DEVICES;
+         tok = strtok_r(NULL, ",", &save), i++)
+      snprintf(g_card_path[i], sizeof(g_card_path[i]),
+               "/sys/class/drm/%s/device/gpu_busy_percent", tok);
+    return;
+  }
+  /* fall back to enumeration order of card*/
+  int idx = 0;
+  for (int c = 0; c < 64 && idx < VGPU_MAX_DEVICES; c++) {
+    char p[256];
+    snprintf(p, sizeof(p), "/sys/class/drm/card%d/device/gpu_busy_percent", c);
+    if (access(p, R_OK) == 0)
+      snprintf(g_card_path[idx++], sizeof(g_card_path[0]), "%s", p);
+  }
+}
+
+static int read_busy_percent(int dev) {
+  const char *util_file = getenv("VGPU_UTIL_FILE");
+  char buf[32] = {0};
+  int fd = -1;
+  if (util_file && *util_file) {
+    fd = open(util_file, O_RDONLY);
+  } else {
+    resolve_cards();
+    if (dev < 0 || dev >= VGPU_MAX_DEVICES || !g_card_path[dev][0]) return -1;
+    fd = open(g_card_path[dev], O_RDONLY);
+  }
+  if (fd < 0) return -1;
+  ssize_t n = read(fd, buf, sizeof(buf) - 1);
+  close(fd);
+  if (n <= 0) return -1;
+  return atoi(buf);
+}
+
+static double fixed_rate(void) {
+  const char *e = getenv("VGPU_TOKEN_RATE");
+  return e ? atof(e) : 0.0;
+}
+
+static void refill(vgpu_region_t *r, uint64_t now) {
+  uint64_t last = __atomic_load_n(&r->last_refill_ns, __ATOMIC_RELAXED);
+  if (now - last < REFILL_INTERVAL_NS) return;
+  if (!__atomic_compare_exchange_n(&r->last_refill_ns, &last, now, 0,
+                                   __ATOMIC_ACQ_REL, __ATOMIC_RELAXED))
+    return; /* another process holds this tick's lease */
+  double dt = (double)(now - last) / NSEC;
+  if (dt > 1.0) dt = 1.0;
+  double fixed = fixed_rate();
+  for (int d = 0; d < VGPU_MAX_DEVICES; d++) {
+    uint64_t lim = r->sm_limit[d];
+    if (lim == 0 || lim >= 100) continue;
+    double base = fixed > 0 ? fixed : RATE_FULL * (double)lim / 100.0;
+    if (fixed <= 0) {
+      /* utilization feedback: converge measured busy% on the limit */
+      int util = read_busy_percent(d);
+      if (util >= 0) {
+        if (g_rate_scale[d] == 0) g_rate_scale[d] = 1.0;
+        if ((uint64_t)util > lim)
+          g_rate_scale[d] *= 0.85;
+        else if ((uint64_t)util < lim * 9 / 10)
+          g_rate_scale[d] *= 1.08;
+        if (g_rate_scale[d] < 0.02) g_rate_scale[d] = 0.02;
+        if (g_rate_scale[d] > 50.0) g_rate_scale[d] = 50.0;
+        base *= g_rate_scale[d];
+      }
+    }
+    int64_t add = (int64_t)(base * dt);
+    int64_t cap = (int64_t)(base * BUCKET_SECONDS);
+    if (cap < 1) cap = 1;
+    int64_t cur = __atomic_load_n(&r->core_tokens[d], __ATOMIC_RELAXED);
+    int64_t next = cur + add;
+    if (next > cap) next = cap;
+    __atomic_store_n(&r->core_tokens[d], next, __ATOMIC_RELAXED);
+    __atomic_store_n(&r->token_fill_rate[d], (int64_t)base, __ATOMIC_RELAXED);
+  }
+}
+
+static void *watcher_main(void *arg) {
+  (void)arg;
+  vgpu_region_t *r = vgpu_region_get();
+  if (!r) return NULL;
+  for (;;) {
+    refill(r, now_ns());
+    usleep(25000);
+  }
+  return NULL;
+}
+
+void vgpu_limiter_init(void) {
+  pthread_mutex_lock(&g_watch_mu);
+  if (!g_watcher_started) {
+    vgpu_region_t *r = vgpu_region_get();
+    int need = 0;
+    if (r)
+      for (int d = 0; d < VGPU_MAX_DEVICES; d++)
+        if (r->sm_limit[d] > 0 && r->sm_limit[d] < 100) need = 1;
+    const char *policy = getenv(ENV_CORE_POLICY);
+    if (policy && strcasecmp(policy, "disable") == 0) need = 0;
+    if (need && pthread_create(&g_watcher, NULL, watcher_main, NULL) == 0) {
+      pthread_detach(g_watcher);
+      g_watcher_started = 1;
+      vgpu_log(VGPU_INFO, "CU limiter watcher started (limit0=%llu%%)",
+               (unsigned long long)(r ? r->sm_limit[0] : 0));
+    }
+  }
+  pthread_mutex_unlock(&g_watch_mu);
+}
+
+void vgpu_limiter_gate(int dev, uint64_t workgroups) {
+  vgpu_region_t *r = vgpu_region_get();
+  if (!r || dev < 0 || dev >= VGPU_MAX_DEVICES) return;
+
+  /* priority gate: the monitor writes recent_kernel = -1 to suspend
+   * lower-priority containers while a high-priority one is active
+   * (reference feedback.go:197-255) */
+  int waited_ms = 0;
+  while (__atomic_load_n(&r->recent_kernel, __ATOMIC_RELAXED) < 0 &&
+         waited_ms < 60000) {
+    usleep(1000);
+    waited_ms++;
+  }
+  int32_t rk = __atomic_load_n(&r->recent_kernel, __ATOMIC_RELAXED);
+  if (rk < 100) __atomic_store_n(&r->recent_kernel, rk + 1, __ATOMIC_RELAXED);
+
+  uint64_t lim = r->sm_limit[dev];
+  if (lim == 0 || lim >= 100) return;
+  const char *policy = getenv(ENV_CORE_POLICY);
+  if (policy && strcasecmp(policy, "disable") == 0) return;
+  if (__atomic_load_n(&r->utilization_switch, __ATOMIC_RELAXED) == 0)
+    return; /* monitor says: uncontended, free-run */
+  if (!g_watcher_started) vgpu_limiter_init();
+
+  int64_t cost = (int64_t)workgroups;
+  if (cost < 1) cost = 1;
+  if (cost > 65536) cost = 65536; /* one giant grid shouldn't starve forever */
+  for (;;) {
+    int64_t cur = __atomic_load_n(&r->core_tokens[dev], __ATOMIC_RELAXED);
+    if (cur > 0) {
+      __atomic_fetch_sub(&r->core_tokens[dev], cost, __ATOMIC_RELAXED);
+      return;
+    }
+    refill(r, now_ns()); /* self-refill if the watcher lease is idle */
+    usleep(200);
+  }
+}
+
+/* ---- launch hooks ---------------------------------------------------- */
+typedef struct { unsigned x, y, z; } vdim3;
+
+hipError_t hipLaunchKernel(const void *f, vdim3 grid, vdim3 block, void **args,
+                           size_t shared, void *stream) {
+  typedef hipError_t (*fn)(const void *, vdim3, vdim3, void **, size_t, void *);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipLaunchKernel");
+  if (!real) return hipErrorInvalidValue;
+  vgpu_ensure_initialized();
+  if (!vgpu_control_disabled())
+    vgpu_limiter_gate(vgpu_current_device(),
+                      (uint64_t)grid.x * grid.y * grid.z);
+  return real(f, grid, block, args, shared, stream);
+}
+
+hipError_t hipModuleLaunchKernel(void *func, unsigned gx, unsigned gy,
+                                 unsigned gz, unsigned bx, unsigned by,
+                                 unsigned bz, unsigned sharedMem, void *stream,
+                                 void **params, void **extra) {
+  typedef hipError_t (*fn)(void *, unsigned, unsigned, unsigned, unsigned,
+                           unsigned, unsigned, unsigned, void *, void **,
+                           void **);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipModuleLaunchKernel");
+  if (!real) return hipErrorInvalidValue;
+  vgpu_ensure_initialized();
+  if (!vgpu_control_disabled())
+    vgpu_limiter_gate(vgpu_current_device(), (uint64_t)gx * gy * gz);
+  return real(func, gx, gy, gz, bx, by, bz, sharedMem, stream, params, extra);
+}
+
+hipError_t hipExtModuleLaunchKernel(void *func, unsigned gwx, unsigned gwy,
+                                    unsigned gwz, unsigned bx, unsigned by,
+                                    unsigned bz, size_t sharedMem, void *stream,
+                                    void *startEvent, void *stopEvent,
+                                    unsigned flags) {
+  typedef hipError_t (*fn)(void *, unsigned, unsigned, unsigned, unsigned,
+                           unsigned, unsigned, size_t, void *, void *, void *,
+                           unsigned);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipExtModuleLaunchKernel");
+  if (!real) return hipErrorInvalidValue;
+  vgpu_ensure_initialized();
+  if (!vgpu_control_disabled()) {
+    /* ext-launch global sizes are in WORK-ITEMS; convert to workgroups */
+    uint64_t wgs = ((uint64_t)gwx / (bx ? bx : 1)) *
+                   ((uint64_t)gwy / (by ? by : 1)) *
+                   ((uint64_t)gwz / (bz ? bz : 1));
+    vgpu_limiter_gate(vgpu_current_device(), wgs ? wgs : 1);
+  }
+  return real(func, gwx, gwy, gwz, bx, by, bz, sharedMem, stream, startEvent,
+              stopEvent, flags);
+}
+
+hipError_t hipLaunchCooperativeKernel(const void *f, vdim3 grid, vdim3 block,
+                                      void **args, unsigned shared,
+                                      void *stream) {
+  typedef hipError_t (*fn)(const void *, vdim3, vdim3, void **, unsigned,
+                           void *);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipLaunchCooperativeKernel");
+  if (!real) return hipErrorInvalidValue;
+  vgpu_ensure_initialized();
+  if (!vgpu_control_disabled())
+    vgpu_limiter_gate(vgpu_current_device(),
+                      (uint64_t)grid.x * grid.y * grid.z);
+  return real(f, grid, block, args, shared, stream);
+}
+
+hipError_t hipGraphLaunch(void *graphExec, void *stream) {
+  typedef hipError_t (*fn)(void *, void *);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipGraphLaunch");
+  if (!real) return hipErrorInvalidValue;
+  vgpu_ensure_initialized();
+  /* a replayed graph bundles many launches; charge a graph-sized constant
+   * (the per-node hooks do not fire on replay) */
+  if (!vgpu_control_disabled())
+    vgpu_limiter_gate(vgpu_current_device(), 2048);
+  return real(graphExec, stream);
+}

@@ -1,0 +1,228 @@
+"""libvgpu-hip interceptor tests against the fake HIP runtime.
+
+True LD_PRELOAD interposition: hip_consumer is PLT-linked against SONAME
+libamdhip64.so, LD_LIBRARY_PATH resolves the fake runtime, LD_PRELOAD
+injects the interceptor — the exact in-container mechanics, no GPU needed
+(the fake-vendor-library pattern of the reference's cndev mock,
+/root/reference/pkg/device-plugin/mlu/cndev/mock/cndev.c).
+"""
+import ctypes
+import json
+import os
+import subprocess
+import time
+from pathlib import Path
+
+import pytest
+
+CSRC = Path(__file__).resolve().parent.parent / "k8s_device_plugin_amd" / "csrc"
+LIBVGPU = CSRC / "libvgpu-hip.so"
+FAKEDIR = CSRC / "fakehip"
+CONSUMER = CSRC / "test" / "hip_consumer"
+
+MIB = 1024 * 1024
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    if not (LIBVGPU.exists() and CONSUMER.exists()):
+        subprocess.run(
+            ["make", "libvgpu-hip.so", "fakehip/libamdhip64.so", "test/hip_consumer"],
+            cwd=CSRC, check=True, capture_output=True,
+        )
+
+
+def run_consumer(args, cache, mem_limit=None, extra_env=None, preload=True):
+    env = dict(os.environ)
+    env["LD_LIBRARY_PATH"] = str(FAKEDIR)
+    if preload:
+        env["LD_PRELOAD"] = str(LIBVGPU)
+    env["VGPU_DEVICE_MEMORY_SHARED_CACHE"] = str(cache)
+    env["VGPU_REAL_HIP_PATH"] = str(FAKEDIR / "libamdhip64.so")
+    if mem_limit is not None:
+        env["VGPU_DEVICE_MEMORY_LIMIT"] = mem_limit
+    env.update(extra_env or {})
+    out = subprocess.run(
+        [str(CONSUMER)] + [str(a) for a in args],
+        env=env, capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stderr
+    return [json.loads(line) for line in out.stdout.splitlines()]
+
+
+class TestMemoryCap:
+    def test_cap_enforced(self, tmp_path):
+        res = run_consumer(
+            ["meminfo", "alloc", 600 * MIB, "alloc", 600 * MIB, "meminfo"],
+            tmp_path / "r.cache", mem_limit="1000m",
+        )
+        assert res[0] == {"cmd": "meminfo", "free": 1000 * MIB, "total": 1000 * MIB, "err": 0}
+        assert res[1]["err"] == 0
+        assert res[2]["err"] == 2  # hipErrorOutOfMemory
+        assert res[3]["free"] == 400 * MIB
+
+    def test_free_credits_back(self, tmp_path):
+        res = run_consumer(
+            ["alloc", 600 * MIB, "free", "alloc", 900 * MIB, "meminfo"],
+            tmp_path / "r.cache", mem_limit="1000m",
+        )
+        assert [r["err"] for r in res[:3]] == [0, 0, 0]
+        assert res[3]["free"] == 100 * MIB
+
+    def test_72g_quota_of_288g(self, tmp_path):
+        """BASELINE config 2 shape: 4-way split of 288 GB -> 72 GB caps."""
+        res = run_consumer(
+            ["totalmem", "meminfo"], tmp_path / "r.cache", mem_limit="73728m",
+        )
+        assert res[0]["total"] == 73728 * MIB
+        assert res[1]["total"] == 73728 * MIB
+
+    def test_gigabyte_suffix(self, tmp_path):
+        res = run_consumer(["totalmem"], tmp_path / "r.cache", mem_limit="72g")
+        assert res[0]["total"] == 72 * 1024 * MIB
+
+    def test_no_limit_passthrough(self, tmp_path):
+        res = run_consumer(["meminfo"], tmp_path / "r.cache")
+        assert res[0]["total"] == 288 * 1024 * MIB  # fake default = MI355X HBM
+
+    def test_disable_control(self, tmp_path):
+        res = run_consumer(
+            ["alloc", 600 * MIB, "alloc", 600 * MIB],
+            tmp_path / "r.cache", mem_limit="1000m",
+            extra_env={"VGPU_DISABLE_CONTROL": "1"},
+        )
+        assert [r["err"] for r in res] == [0, 0]
+
+    def test_per_device_limit_env(self, tmp_path):
+        res = run_consumer(
+            ["totalmem"], tmp_path / "r.cache",
+            extra_env={"VGPU_DEVICE_MEMORY_LIMIT_0": "500m"},
+        )
+        assert res[0]["total"] == 500 * MIB
+
+
+class TestMultiProcessAccounting:
+    def test_usage_shared_across_processes(self, tmp_path):
+        """Two processes share one region: the second sees the first's usage.
+
+        Reference analog: multiprocess_memory_limit shared-region ledger
+        (SURVEY.md §2.6).
+        """
+        cache = tmp_path / "r.cache"
+        env = dict(os.environ)
+        env.update({
+            "LD_LIBRARY_PATH": str(FAKEDIR),
+            "LD_PRELOAD": str(LIBVGPU),
+            "VGPU_DEVICE_MEMORY_SHARED_CACHE": str(cache),
+            "VGPU_REAL_HIP_PATH": str(FAKEDIR / "libamdhip64.so"),
+            "VGPU_DEVICE_MEMORY_LIMIT": "1000m",
+        })
+        # proc A holds 700 MiB and sleeps; proc B can only get 200 MiB
+        a = subprocess.Popen(
+            [str(CONSUMER), "alloc", str(700 * MIB), "sleep", "6000"],
+            env=env, stdout=subprocess.PIPE, text=True,
+        )
+        try:
+            deadline = time.time() + 5
+            line = a.stdout.readline()
+            assert json.loads(line)["err"] == 0
+            res = run_consumer(
+                ["alloc", str(500 * MIB), "alloc", str(200 * MIB)],
+                cache, mem_limit="1000m",
+            )
+            assert res[0]["err"] == 2  # 700 + 500 > 1000
+            assert res[1]["err"] == 0  # 700 + 200 fits
+            assert time.time() < deadline + 60
+        finally:
+            a.kill()
+            a.wait()
+
+    def test_dead_process_usage_reclaimed(self, tmp_path):
+        cache = tmp_path / "r.cache"
+        # proc A allocates 900 MiB and EXITS (killed) without freeing
+        res = run_consumer(["alloc", str(900 * MIB)], cache, mem_limit="1000m")
+        assert res[0]["err"] == 0
+        # its pid is gone; a new proc must see the slot pruned
+        res = run_consumer(["alloc", str(900 * MIB)], cache, mem_limit="1000m")
+        assert res[0]["err"] == 0
+
+
+class TestOversubscription:
+    def test_oversubscribe_uses_managed(self, tmp_path):
+        """BASELINE config 5 shape: limit above physical; allocs become
+        managed so XNACK can page to host DRAM."""
+        res = run_consumer(
+            ["alloc", str(100 * MIB), "stats"], tmp_path / "r.cache",
+            mem_limit="409600m",  # 400 GB virtual on a 288 GB card
+            extra_env={"VGPU_OVERSUBSCRIBE": "true"},
+        )
+        assert res[0]["err"] == 0
+        assert res[1]["managed"] == 1
+
+    def test_no_oversubscribe_no_managed(self, tmp_path):
+        res = run_consumer(
+            ["alloc", str(100 * MIB), "stats"], tmp_path / "r.cache",
+            mem_limit="1000m",
+        )
+        assert res[1]["managed"] == 0
+
+
+class TestLimiter:
+    def test_token_bucket_paces_launches(self, tmp_path):
+        """With a fixed token rate, 100 launches of 5000 workgroups cost
+        500k tokens -> >= ~0.4 s at 1M tokens/s.  Without a CU limit the
+        same storm is instant."""
+        t0 = time.time()
+        res = run_consumer(
+            ["launch", 100, 5000], tmp_path / "a.cache",
+            extra_env={"VGPU_DEVICE_CU_LIMIT": "10", "VGPU_TOKEN_RATE": "1000000"},
+        )
+        paced = res[0]["seconds"]
+        res2 = run_consumer(["launch", 100, 5000], tmp_path / "b.cache")
+        free = res2[0]["seconds"]
+        assert res[0]["err"] == 0 and res2[0]["err"] == 0
+        assert paced > 0.25, f"throttled storm finished too fast: {paced}"
+        assert free < 0.2, f"unthrottled storm too slow: {free}"
+        assert time.time() - t0 < 60
+
+    def test_core_policy_disable(self, tmp_path):
+        res = run_consumer(
+            ["launch", 100, 5000], tmp_path / "r.cache",
+            extra_env={
+                "VGPU_DEVICE_CU_LIMIT": "10",
+                "VGPU_TOKEN_RATE": "1000000",
+                "GPU_CORE_UTILIZATION_POLICY": "disable",
+            },
+        )
+        assert res[0]["seconds"] < 0.2
+
+    def test_limit_100_not_throttled(self, tmp_path):
+        res = run_consumer(
+            ["launch", 100, 5000], tmp_path / "r.cache",
+            extra_env={"VGPU_DEVICE_CU_LIMIT": "100", "VGPU_TOKEN_RATE": "1000000"},
+        )
+        assert res[0]["seconds"] < 0.2
+
+
+class TestRegionABI:
+    def test_layout_json_parses(self):
+        lib = ctypes.CDLL(str(LIBVGPU))
+        buf = ctypes.create_string_buffer(4096)
+        n = lib.vgpu_region_layout_json(buf, 4096)
+        assert 0 < n < 4096
+        layout = json.loads(buf.value.decode())
+        assert layout["_max_devices"] == 16
+        assert layout["_max_procs"] == 1024
+        assert layout["_size"] > 0
+        # monitor-feedback fields must exist and be word-aligned
+        for f in ("recent_kernel", "utilization_switch", "priority", "procs"):
+            assert layout[f] % 4 == 0
+
+    def test_region_file_created_with_layout_size(self, tmp_path):
+        cache = tmp_path / "r.cache"
+        run_consumer(["meminfo"], cache, mem_limit="100m")
+        lib = ctypes.CDLL(str(LIBVGPU))
+        buf = ctypes.create_string_buffer(4096)
+        lib.vgpu_region_layout_json(buf, 4096)
+        layout = json.loads(buf.value.decode())
+        assert cache.stat().st_size == layout["_size"]
