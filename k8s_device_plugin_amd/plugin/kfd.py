@@ -1,0 +1,154 @@
+"""MI355X enumeration from KFD sysfs.
+
+No cgo, no vendor CLI parsing: everything the plugin needs is in
+``/sys/class/kfd/kfd/topology/nodes/*`` (the reference's DCU plugin shells
+out to hy-smi/hdmcli and parses with Sscanf — server.go:50-175; the NVIDIA
+path uses NVML.  KFD sysfs is the MI355X-native source):
+
+- ``properties``: simd_count / simd_per_cu -> CU count (MI355X: 1024/4=256),
+  unique_id -> stable UUID, location_id+domain -> PCI BDF, drm_render_minor
+  -> /dev/dri node mapping, gfx_target_version.
+- ``mem_banks/*/properties``: size_in_bytes -> HBM capacity (288 GB).
+- NUMA: /sys/bus/pci/devices/<bdf>/numa_node (no hwloc needed — the
+  reference itself reads sysfs on the NVIDIA path, rm/nvml_devices.go:134).
+
+The sysfs roots are constructor parameters so tests run against fixture
+trees (the cndev-mock testing pattern).
+"""
+from __future__ import annotations
+
+import glob
+import logging
+import os
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+log = logging.getLogger(__name__)
+
+KFD_TOPOLOGY = "/sys/class/kfd/kfd/topology"
+PCI_DEVICES = "/sys/bus/pci/devices"
+KFD_DEV = "/dev/kfd"
+
+
+@dataclass
+class PhysicalGPU:
+    index: int              # enumeration order (stable: by node id)
+    node_id: int            # KFD topology node number
+    gpu_id: int             # KFD gpu_id
+    uuid: str               # "GPU-<unique_id hex>"
+    cu_count: int           # 256 on MI355X
+    mem_bytes: int          # HBM3E bank size (288 GB)
+    numa_node: int
+    pci_bdf: str            # "0000:0c:00.0"
+    drm_render_minor: int   # /dev/dri/renderD<minor>
+    gfx_target: str         # e.g. "gfx950"
+    io_links: Dict[int, int] = field(default_factory=dict)  # node_to -> type
+
+    @property
+    def drm_card(self) -> int:
+        return self.drm_render_minor - 128
+
+    @property
+    def device_paths(self) -> List[str]:
+        return [
+            KFD_DEV,
+            f"/dev/dri/card{self.drm_card}",
+            f"/dev/dri/renderD{self.drm_render_minor}",
+        ]
+
+
+def _read_properties(path: str) -> Dict[str, int]:
+    props: Dict[str, int] = {}
+    try:
+        with open(path) as f:
+            for line in f:
+                parts = line.split()
+                if len(parts) == 2:
+                    try:
+                        props[parts[0]] = int(parts[1])
+                    except ValueError:
+                        pass
+    except OSError:
+        pass
+    return props
+
+
+def _read_int(path: str, default: int = -1) -> int:
+    try:
+        with open(path) as f:
+            return int(f.read().strip(), 0)
+    except (OSError, ValueError):
+        return default
+
+
+def enumerate_gpus(
+    topology_root: str = KFD_TOPOLOGY, pci_root: str = PCI_DEVICES
+) -> List[PhysicalGPU]:
+    gpus: List[PhysicalGPU] = []
+    nodes_dir = os.path.join(topology_root, "nodes")
+    if not os.path.isdir(nodes_dir):
+        log.warning("KFD topology not present at %s", nodes_dir)
+        return gpus
+    for node_path in sorted(
+        glob.glob(os.path.join(nodes_dir, "*")),
+        key=lambda p: int(os.path.basename(p)) if os.path.basename(p).isdigit() else 1 << 30,
+    ):
+        node_name = os.path.basename(node_path)
+        if not node_name.isdigit():
+            continue
+        props = _read_properties(os.path.join(node_path, "properties"))
+        simd_count = props.get("simd_count", 0)
+        if simd_count <= 0:
+            continue  # CPU node
+        simd_per_cu = props.get("simd_per_cu", 4) or 4
+        cu_count = simd_count // simd_per_cu
+        gpu_id = _read_int(os.path.join(node_path, "gpu_id"), 0)
+        unique_id = props.get("unique_id", 0)
+        uuid = f"GPU-{unique_id:016x}" if unique_id else f"GPU-kfd-{gpu_id}"
+        location = props.get("location_id", 0)
+        domain = props.get("domain", 0)
+        bdf = f"{domain:04x}:{(location >> 8) & 0xff:02x}:{(location >> 3) & 0x1f:02x}.{location & 0x7}"
+        mem_bytes = 0
+        for bank in glob.glob(os.path.join(node_path, "mem_banks", "*", "properties")):
+            bprops = _read_properties(bank)
+            # heap_type 1/2 = FB public/private (device HBM)
+            if bprops.get("heap_type", 0) in (1, 2):
+                mem_bytes += bprops.get("size_in_bytes", 0)
+        numa = _read_int(os.path.join(pci_root, bdf, "numa_node"), -1)
+        if numa < 0:
+            numa = 0
+        gfx_ver = props.get("gfx_target_version", 0)
+        # 90500 -> gfx950 encoding: major*10000 + minor*100 + step
+        gfx = f"gfx{gfx_ver // 10000}{(gfx_ver // 100) % 100}{gfx_ver % 100:x}" if gfx_ver else ""
+        io_links: Dict[int, int] = {}
+        for link in glob.glob(os.path.join(node_path, "io_links", "*", "properties")):
+            lprops = _read_properties(link)
+            if "node_to" in lprops:
+                io_links[lprops["node_to"]] = lprops.get("type", 0)
+        gpus.append(
+            PhysicalGPU(
+                index=len(gpus),
+                node_id=int(node_name),
+                gpu_id=gpu_id,
+                uuid=uuid,
+                cu_count=cu_count,
+                mem_bytes=mem_bytes,
+                numa_node=numa,
+                pci_bdf=bdf,
+                drm_render_minor=props.get("drm_render_minor", 128 + len(gpus)),
+                gfx_target=gfx,
+                io_links=io_links,
+            )
+        )
+    return gpus
+
+
+def kfd_healthy(kfd_dev: str = KFD_DEV) -> bool:
+    """Health = /dev/kfd openable (reference DCU simpleHealthCheck,
+    dcu/server.go:225-234); per-device RAS checks layer on top."""
+    try:
+        fd = os.open(kfd_dev, os.O_RDONLY)
+        os.close(fd)
+        return True
+    except OSError:
+        return False

@@ -1,0 +1,88 @@
+"""Resource manager: physical GPU inventory -> fake-device fan-out.
+
+Reference: pkg/device-plugin/.../rm/devices.go:144-166 — each physical GPU
+becomes ``device_split_count`` kubelet devices with IDs ``<UUID>-<i>``, and
+memory/cores scaling is applied when advertising (register.go:96-162).
+"""
+from __future__ import annotations
+
+import logging
+from dataclasses import dataclass
+from typing import Dict, List, Optional
+
+from .. import MI355X_DEVICE_TYPE
+from ..utils.types import DeviceInfo
+from .kfd import PhysicalGPU
+
+log = logging.getLogger(__name__)
+
+MIB = 1024 * 1024
+
+
+@dataclass
+class FakeDevice:
+    id: str           # "<uuid>-<i>"
+    uuid: str         # physical uuid
+    numa: int
+    healthy: bool = True
+
+
+class ResourceManager:
+    def __init__(self, gpus: List[PhysicalGPU], split_count: int = 10,
+                 memory_scaling: float = 1.0, cores_scaling: float = 1.0,
+                 device_type: str = MI355X_DEVICE_TYPE):
+        self.gpus = gpus
+        self.split_count = max(1, split_count)
+        self.memory_scaling = memory_scaling
+        self.cores_scaling = cores_scaling
+        self.device_type = device_type
+        self.health: Dict[str, bool] = {g.uuid: True for g in gpus}
+
+    def by_uuid(self, uuid: str) -> Optional[PhysicalGPU]:
+        for g in self.gpus:
+            if g.uuid == uuid:
+                return g
+        return None
+
+    def fake_devices(self) -> List[FakeDevice]:
+        out: List[FakeDevice] = []
+        for g in self.gpus:
+            for i in range(self.split_count):
+                out.append(FakeDevice(
+                    id=f"{g.uuid}-{i}",
+                    uuid=g.uuid,
+                    numa=g.numa_node,
+                    healthy=self.health.get(g.uuid, True),
+                ))
+        return out
+
+    @staticmethod
+    def uuid_of_fake(fake_id: str) -> str:
+        """'GPU-abc-3' -> 'GPU-abc' (strip the trailing replica index)."""
+        if "-" in fake_id:
+            head, _, tail = fake_id.rpartition("-")
+            if tail.isdigit():
+                return head
+        return fake_id
+
+    def api_devices(self) -> List[DeviceInfo]:
+        """Node-annotation inventory with scaling applied
+        (reference getApiDevices, register.go:96-162)."""
+        out = []
+        for idx, g in enumerate(self.gpus):
+            out.append(DeviceInfo(
+                id=g.uuid,
+                count=self.split_count,
+                devmem=int(g.mem_bytes / MIB * self.memory_scaling),
+                devcore=int(100 * self.cores_scaling),
+                type=self.device_type,
+                numa=g.numa_node,
+                health=self.health.get(g.uuid, True),
+                index=idx,
+            ))
+        return out
+
+    def set_health(self, uuid: str, healthy: bool) -> bool:
+        old = self.health.get(uuid)
+        self.health[uuid] = healthy
+        return old != healthy
