@@ -1,0 +1,216 @@
+"""MI355X hardware tests: real enforcement through the real ROCm stack.
+
+Every test is @pytest.mark.gpu.  Enforcement checks run in child processes
+with LD_PRELOAD, exactly like a pod container; kernels come from the
+in-tree gfx950 probe library (csrc/vgpu/probe.hip).
+"""
+import ctypes
+import json
+import os
+import subprocess
+import sys
+import time
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).resolve().parent.parent
+CSRC = REPO / "k8s_device_plugin_amd" / "csrc"
+LIBVGPU = CSRC / "libvgpu-hip.so"
+LIBPROBE = CSRC / "libvgpu_probe.so"
+
+GIB = 1 << 30
+
+pytestmark = pytest.mark.gpu
+
+
+def run_child(code, extra_env=None, timeout=240):
+    env = dict(os.environ)
+    env.update(extra_env or {})
+    out = subprocess.run([sys.executable, "-c", code], env=env,
+                         capture_output=True, text=True, timeout=timeout,
+                         cwd=str(REPO))
+    assert out.returncode == 0, f"stderr: {out.stderr[-2000:]}"
+    return json.loads(out.stdout.strip().splitlines()[-1])
+
+
+def preload_env(tmp_path, limit=None, extra=None):
+    env = {
+        "LD_PRELOAD": str(LIBVGPU),
+        "VGPU_DEVICE_MEMORY_SHARED_CACHE": str(tmp_path / "region.cache"),
+    }
+    if limit:
+        env["VGPU_DEVICE_MEMORY_LIMIT"] = limit
+    env.update(extra or {})
+    return env
+
+
+class TestProbeKernels:
+    def test_vecadd_numerics(self):
+        probe = ctypes.CDLL(str(LIBPROBE))
+        assert probe.vgpu_probe_vecadd(ctypes.c_size_t(1 << 22)) == 0
+
+    def test_vecadd_numerics_under_preload(self, tmp_path):
+        code = (
+            "import ctypes, json;"
+            f"p = ctypes.CDLL('{LIBPROBE}');"
+            "rc = p.vgpu_probe_vecadd(ctypes.c_size_t(1 << 22));"
+            "print(json.dumps({'rc': rc}))"
+        )
+        res = run_child(code, preload_env(tmp_path, limit="8192m"))
+        assert res["rc"] == 0
+
+
+class TestMemoryCapReal:
+    def test_torch_sees_quota_and_ooms(self, tmp_path):
+        """BASELINE config 2: a 72 GB vGPU on the 288 GB card — PyTorch must
+        see 72 GB total and fail allocations beyond it."""
+        code = (
+            "import torch, json; torch.cuda.init();"
+            "free, total = torch.cuda.mem_get_info();"
+            "err = 'none'\n"
+            "bufs = []\n"
+            "try:\n"
+            "    for _ in range(40):\n"
+            "        bufs.append(torch.empty(2 * (1<<30), dtype=torch.uint8, device='cuda'))\n"
+            "except torch.cuda.OutOfMemoryError: err = 'oom'\n"
+            "print(json.dumps({'total': total, 'free': free, 'err': err,"
+            " 'held': len(bufs)}))"
+        )
+        res = run_child(code, preload_env(tmp_path, limit="73728m"), timeout=600)
+        assert res["total"] == 73728 * (1 << 20)
+        assert res["err"] == "oom"
+        # ~72 GiB in 2 GiB chunks minus torch overhead
+        assert 30 <= res["held"] <= 36
+
+    def test_hip_consumer_on_real_runtime(self, tmp_path):
+        out = subprocess.run(
+            [str(CSRC / "test" / "hip_consumer"), "meminfo",
+             "alloc", str(1 * GIB), "alloc", str(2 * GIB), "meminfo"],
+            env={**os.environ, **preload_env(tmp_path, limit="2048m")},
+            capture_output=True, text=True, timeout=240)
+        assert out.returncode == 0, out.stderr
+        lines = [json.loads(l) for l in out.stdout.splitlines()]
+        assert lines[0]["total"] == 2048 * (1 << 20)
+        assert lines[1]["err"] == 0
+        assert lines[2]["err"] == 2  # over quota
+
+    def test_multiprocess_shared_cap(self, tmp_path):
+        """Two processes in one 'container' (same region) share the cap."""
+        env = {**os.environ, **preload_env(tmp_path, limit="4096m")}
+        hold = subprocess.Popen(
+            [str(CSRC / "test" / "hip_consumer"), "alloc", str(3 * GIB),
+             "sleep", "15000"],
+            env=env, stdout=subprocess.PIPE, text=True)
+        try:
+            first = json.loads(hold.stdout.readline())
+            assert first["err"] == 0
+            out = subprocess.run(
+                [str(CSRC / "test" / "hip_consumer"), "alloc", str(2 * GIB),
+                 "alloc", str(512 * (1 << 20))],
+                env=env, capture_output=True, text=True, timeout=240)
+            lines = [json.loads(l) for l in out.stdout.splitlines()]
+            assert lines[0]["err"] == 2  # 3G + 2G > 4G
+            assert lines[1]["err"] == 0  # 3G + 0.5G fits
+        finally:
+            hold.kill()
+            hold.wait()
+
+
+class TestOversubscriptionReal:
+    def test_400g_virtual_on_288g(self, tmp_path):
+        """BASELINE config 5: 400 GB quota on 288 GB HBM; allocations route
+        through hipMallocManaged (XNACK) and the tools see 400 GB."""
+        code = (
+            "import torch, json; torch.cuda.init();"
+            "free, total = torch.cuda.mem_get_info();"
+            "x = torch.empty(1 << 30, dtype=torch.uint8, device='cuda');"
+            "x[:] = 1;"
+            "print(json.dumps({'total': total, 'sum': int(x[:10].sum())}))"
+        )
+        res = run_child(code, preload_env(
+            tmp_path, limit="409600m",
+            extra={"VGPU_OVERSUBSCRIBE": "true", "HSA_XNACK": "1"}),
+            timeout=600)
+        assert res["total"] == 409600 * (1 << 20)
+        assert res["sum"] == 10
+
+
+class TestCUMaskReal:
+    def test_hsa_cu_mask_partitions(self, tmp_path):
+        """HSA_CU_MASK 0:0-63 (2 XCDs of 8) must slow a chip-filling burn by
+        ~4x vs all 256 CUs — the hard partition works end to end."""
+        code = (
+            "import ctypes, json;"
+            f"p = ctypes.CDLL('{LIBPROBE}');"
+            "p.vgpu_probe_burn.restype = ctypes.c_double;"
+            "t = p.vgpu_probe_burn(4, 256, 20);"
+            "print(json.dumps({'t': t}))"
+        )
+        full = run_child(code, {})
+        masked = run_child(code, {"HSA_CU_MASK": "0:0-63"})
+        assert full["t"] > 0 and masked["t"] > 0
+        ratio = masked["t"] / full["t"]
+        assert ratio > 2.0, f"CU mask had no effect: ratio {ratio:.2f}"
+
+
+class TestLimiterReal:
+    def test_token_bucket_paces_real_storm(self, tmp_path):
+        code = (
+            "import ctypes, json;"
+            f"p = ctypes.CDLL('{LIBPROBE}');"
+            "p.vgpu_probe_storm.restype = ctypes.c_double;"
+            "t = p.vgpu_probe_storm(500, 512);"
+            "print(json.dumps({'t': t}))"
+        )
+        free = run_child(code, preload_env(tmp_path / "a"))
+        (tmp_path / "a").mkdir(exist_ok=True)
+        (tmp_path / "b").mkdir(exist_ok=True)
+        paced = run_child(code, preload_env(
+            tmp_path / "b",
+            extra={"VGPU_DEVICE_CU_LIMIT": "10",
+                   "VGPU_TOKEN_RATE": "200000"}))
+        # 500 launches x 512 wg = 256k tokens at 200k/s => >= ~1 s
+        assert paced["t"] > max(4 * free["t"], 0.8), (
+            f"throttle ineffective: free={free['t']:.3f} paced={paced['t']:.3f}")
+
+    def test_utilization_feedback_limits_busy(self, tmp_path):
+        """Feedback mode (no fixed rate): a 25% CU limit must hold
+        gpu_busy_percent well under an unthrottled burn."""
+        (tmp_path / "r").mkdir(exist_ok=True)
+        code = (
+            "import ctypes, json;"
+            f"p = ctypes.CDLL('{LIBPROBE}');"
+            "p.vgpu_probe_burn.restype = ctypes.c_double;"
+            "t = p.vgpu_probe_burn(200, 512, 5);"
+            "print(json.dumps({'t': t}))"
+        )
+        t0 = time.time()
+        free = run_child(code, {})
+        paced = run_child(code, preload_env(
+            tmp_path / "r", extra={"VGPU_DEVICE_CU_LIMIT": "25"}),
+            timeout=600)
+        # a 25% limit should stretch wall time at least ~2x
+        assert paced["t"] > 1.8 * free["t"], (
+            f"feedback throttle weak: free={free['t']:.2f} paced={paced['t']:.2f}")
+        assert time.time() - t0 < 500
+
+
+class TestSMISpoof:
+    def test_rsmi_reports_quota(self, tmp_path):
+        """rocm-smi's library path: dlopen(librocm_smi64) is redirected and
+        memory getters show the quota."""
+        code = (
+            "import ctypes, json;"
+            "lib = ctypes.CDLL('librocm_smi64.so');"
+            "lib.rsmi_init(0);"
+            "total = ctypes.c_uint64(0); used = ctypes.c_uint64(0);"
+            "rc1 = lib.rsmi_dev_memory_total_get(0, 0, ctypes.byref(total));"
+            "rc2 = lib.rsmi_dev_memory_usage_get(0, 0, ctypes.byref(used));"
+            "print(json.dumps({'rc1': rc1, 'rc2': rc2, 'total': total.value,"
+            " 'used': used.value}))"
+        )
+        res = run_child(code, preload_env(tmp_path, limit="73728m"))
+        assert res["rc1"] == 0 and res["rc2"] == 0
+        assert res["total"] == 73728 * (1 << 20)
+        assert res["used"] < res["total"]
