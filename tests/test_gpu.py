@@ -163,9 +163,9 @@ class TestLimiterReal:
             "t = p.vgpu_probe_storm(500, 512);"
             "print(json.dumps({'t': t}))"
         )
-        free = run_child(code, preload_env(tmp_path / "a"))
         (tmp_path / "a").mkdir(exist_ok=True)
         (tmp_path / "b").mkdir(exist_ok=True)
+        free = run_child(code, preload_env(tmp_path / "a"))
         paced = run_child(code, preload_env(
             tmp_path / "b",
             extra={"VGPU_DEVICE_CU_LIMIT": "10",
