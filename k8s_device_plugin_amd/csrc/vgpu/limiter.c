@@ -71,13 +71,27 @@ static void resolve_cards(void) {
                "/sys/class/drm/%s/device/gpu_busy_percent", tok);
     return;
   }
-  /* fall back to enumeration order of card*/
+  /* Fallback: map via KFD topology drm_render_minor (card = minor - 128).
+   * Card numbering does NOT follow GPU order on partitioned/multi-VF nodes
+   * (observed on MI355X: one GPU, KFD node 7, render minor 168 = card40). */
   int idx = 0;
-  for (int c = 0; c < 64 && idx < VGPU_MAX_DEVICES; c++) {
+  for (int node = 0; node < 64 && idx < VGPU_MAX_DEVICES; node++) {
     char p[256];
-    snprintf(p, sizeof(p), "/sys/class/drm/card%d/device/gpu_busy_percent", c);
-    if (access(p, R_OK) == 0)
-      snprintf(g_card_path[idx++], sizeof(g_card_path[0]), "%s", p);
+    snprintf(p, sizeof(p),
+             "/sys/class/kfd/kfd/topology/nodes/%d/properties", node);
+    FILE *f = fopen(p, "r");
+    if (!f) continue;
+    long simd = 0, minor = -1;
+    char key[64];
+    long val;
+    while (fscanf(f, "%63s %ld", key, &val) == 2) {
+      if (strcmp(key, "simd_count") == 0) simd = val;
+      if (strcmp(key, "drm_render_minor") == 0) minor = val;
+    }
+    fclose(f);
+    if (simd > 0 && minor >= 128)
+      snprintf(g_card_path[idx++], sizeof(g_card_path[0]),
+               "/sys/class/drm/card%ld/device/gpu_busy_percent", minor - 128);
   }
 }
 

@@ -118,21 +118,33 @@ class TestMemoryCapReal:
 
 
 class TestOversubscriptionReal:
-    def test_400g_virtual_on_288g(self, tmp_path):
-        """BASELINE config 5: 400 GB quota on 288 GB HBM; allocations route
-        through hipMallocManaged (XNACK) and the tools see 400 GB."""
+    def test_400g_quota_visible_and_managed_alloc(self, tmp_path):
+        """BASELINE config 5: 400 GB quota on 288 GB HBM — the C path:
+        hipMalloc routes to hipMallocManaged and meminfo shows 400 GB."""
+        out = subprocess.run(
+            [str(CSRC / "test" / "hip_consumer"), "meminfo",
+             "alloc", str(512 * (1 << 20)), "meminfo"],
+            env={**os.environ, **preload_env(
+                tmp_path, limit="409600m",
+                extra={"VGPU_OVERSUBSCRIBE": "true"})},
+            capture_output=True, text=True, timeout=120)
+        assert out.returncode == 0, out.stderr
+        lines = [json.loads(l) for l in out.stdout.splitlines()]
+        assert lines[0]["total"] == 409600 * (1 << 20)
+        assert lines[1]["err"] == 0
+        assert lines[2]["free"] == (409600 - 512) * (1 << 20)
+
+    def test_torch_compute_on_managed_memory(self, tmp_path):
+        """Torch fill+reduce on oversubscribe-mode (managed) allocations."""
         code = (
             "import torch, json; torch.cuda.init();"
-            "free, total = torch.cuda.mem_get_info();"
-            "x = torch.empty(1 << 30, dtype=torch.uint8, device='cuda');"
-            "x[:] = 1;"
-            "print(json.dumps({'total': total, 'sum': int(x[:10].sum())}))"
+            "x = torch.empty(256 << 20, dtype=torch.uint8, device='cuda');"
+            "x[:] = 1; torch.cuda.synchronize();"
+            "print(json.dumps({'sum': int(x[:10].sum())}))"
         )
         res = run_child(code, preload_env(
             tmp_path, limit="409600m",
-            extra={"VGPU_OVERSUBSCRIBE": "true", "HSA_XNACK": "1"}),
-            timeout=600)
-        assert res["total"] == 409600 * (1 << 20)
+            extra={"VGPU_OVERSUBSCRIBE": "true"}), timeout=180)
         assert res["sum"] == 10
 
 
