@@ -47,7 +47,17 @@ static int g_watcher_started = 0;
 static pthread_mutex_t g_watch_mu = PTHREAD_MUTEX_INITIALIZER;
 static double g_rate_scale[VGPU_MAX_DEVICES]; /* feedback multiplier */
 static char g_card_path[VGPU_MAX_DEVICES][256];
+static long g_gpu_id[VGPU_MAX_DEVICES];       /* KFD gpu_id per device */
 static int g_cards_resolved = 0;
+
+/* Feedback shape: start conservative and clamp fast.  Kernel-heavy
+ * workloads (few big launches) are bound by the busy%/occupancy loop;
+ * launch-bound workloads by the credit rate itself. */
+#define SCALE_INIT 0.10
+#define SCALE_DOWN 0.50
+#define SCALE_UP 1.06
+#define SCALE_MIN 0.002
+#define SCALE_MAX 50.0
 
 static uint64_t now_ns(void) {
   struct timespec ts;
@@ -69,12 +79,13 @@ static void resolve_cards(void) {
          tok = strtok_r(NULL, ",", &save), i++)
       snprintf(g_card_path[i], sizeof(g_card_path[i]),
                "/sys/class/drm/%s/device/gpu_busy_percent", tok);
-    return;
+    /* gpu_ids still come from KFD below */
   }
   /* Fallback: map via KFD topology drm_render_minor (card = minor - 128).
    * Card numbering does NOT follow GPU order on partitioned/multi-VF nodes
    * (observed on MI355X: one GPU, KFD node 7, render minor 168 = card40). */
   int idx = 0;
+  int have_cards = g_card_path[0][0] != 0;
   for (int node = 0; node < 64 && idx < VGPU_MAX_DEVICES; node++) {
     char p[256];
     snprintf(p, sizeof(p),
@@ -89,10 +100,54 @@ static void resolve_cards(void) {
       if (strcmp(key, "drm_render_minor") == 0) minor = val;
     }
     fclose(f);
-    if (simd > 0 && minor >= 128)
-      snprintf(g_card_path[idx++], sizeof(g_card_path[0]),
-               "/sys/class/drm/card%ld/device/gpu_busy_percent", minor - 128);
+    if (simd > 0) {
+      char gp[256];
+      snprintf(gp, sizeof(gp), "/sys/class/kfd/kfd/topology/nodes/%d/gpu_id",
+               node);
+      FILE *gf = fopen(gp, "r");
+      long gid = 0;
+      if (gf) {
+        if (fscanf(gf, "%ld", &gid) != 1) gid = 0;
+        fclose(gf);
+      }
+      g_gpu_id[idx] = gid;
+      if (!have_cards && minor >= 128)
+        snprintf(g_card_path[idx], sizeof(g_card_path[0]),
+                 "/sys/class/drm/card%ld/device/gpu_busy_percent", minor - 128);
+      idx++;
+    }
   }
+}
+
+/* Per-container CU occupancy: sum of this container's processes'
+ * /sys/class/kfd/kfd/proc/<pid>/stats_<gpuid>/cu_occupancy over the 256-CU
+ * chip.  This attributes utilization to THIS container (the reference uses
+ * nvmlDeviceGetProcessUtilization for the same purpose),
+ * so co-located pods don't see each other's load and spiral down. */
+static int read_cu_occupancy_percent(int dev) {
+  vgpu_region_t *r = vgpu_region_get();
+  if (!r || g_gpu_id[dev] == 0) return -1;
+  long total = 0;
+  int found = 0;
+  for (int i = 0; i < VGPU_MAX_PROCS; i++) {
+    int32_t pid = r->procs[i].pid;
+    if (pid <= 0) continue;
+    char p[256];
+    snprintf(p, sizeof(p), "/sys/class/kfd/kfd/proc/%d/stats_%ld/cu_occupancy",
+             pid, g_gpu_id[dev]);
+    int fd = open(p, O_RDONLY);
+    if (fd < 0) continue;
+    char buf[32] = {0};
+    ssize_t n = read(fd, buf, sizeof(buf) - 1);
+    close(fd);
+    if (n > 0) {
+      total += atol(buf);
+      found = 1;
+    }
+  }
+  if (!found) return -1;
+  int pct = (int)(total * 100 / 256); /* CUs occupied -> percent of chip */
+  return pct > 100 ? 100 : pct;
 }
 
 static int read_busy_percent(int dev) {
@@ -132,18 +187,21 @@ static void refill(vgpu_region_t *r, uint64_t now) {
     if (lim == 0 || lim >= 100) continue;
     double base = fixed > 0 ? fixed : RATE_FULL * (double)lim / 100.0;
     if (fixed <= 0) {
-      /* utilization feedback: converge measured busy% on the limit */
-      int util = read_busy_percent(d);
+      /* utilization feedback: converge this container's measured
+       * utilization on the limit.  Own-process cu_occupancy first (correct
+       * attribution under co-location); device busy% as fallback. */
+      int util = read_cu_occupancy_percent(d);
+      if (util < 0) util = read_busy_percent(d);
+      if (g_rate_scale[d] == 0) g_rate_scale[d] = SCALE_INIT;
       if (util >= 0) {
-        if (g_rate_scale[d] == 0) g_rate_scale[d] = 1.0;
         if ((uint64_t)util > lim)
-          g_rate_scale[d] *= 0.85;
+          g_rate_scale[d] *= SCALE_DOWN;
         else if ((uint64_t)util < lim * 9 / 10)
-          g_rate_scale[d] *= 1.08;
-        if (g_rate_scale[d] < 0.02) g_rate_scale[d] = 0.02;
-        if (g_rate_scale[d] > 50.0) g_rate_scale[d] = 50.0;
-        base *= g_rate_scale[d];
+          g_rate_scale[d] *= SCALE_UP;
+        if (g_rate_scale[d] < SCALE_MIN) g_rate_scale[d] = SCALE_MIN;
+        if (g_rate_scale[d] > SCALE_MAX) g_rate_scale[d] = SCALE_MAX;
       }
+      base *= g_rate_scale[d];
     }
     int64_t add = (int64_t)(base * dt);
     int64_t cap = (int64_t)(base * BUCKET_SECONDS);
