@@ -1,0 +1,194 @@
+"""Monitor tests: region mirror, feedback blocking, path GC, metrics.
+
+The regions under test are REAL ones created by the C interceptor in
+consumer processes (fake HIP runtime) — the Python mirror is validated
+against the C writer, not against itself.
+"""
+import json
+import os
+import subprocess
+import threading
+import time
+from pathlib import Path
+
+import pytest
+
+from k8s_device_plugin_amd.monitor.feedback import FeedbackLoop
+from k8s_device_plugin_amd.monitor.metrics import MonitorCollector, metrics_text
+from k8s_device_plugin_amd.monitor.pathmon import GC_GRACE_SECONDS, PathMonitor
+from k8s_device_plugin_amd.monitor.region import SharedRegion, region_layout
+
+CSRC = Path(__file__).resolve().parent.parent / "k8s_device_plugin_amd" / "csrc"
+LIBVGPU = CSRC / "libvgpu-hip.so"
+FAKEDIR = CSRC / "fakehip"
+CONSUMER = CSRC / "test" / "hip_consumer"
+MIB = 1024 * 1024
+
+
+def consumer_env(cache, limit="1000m", extra=None):
+    env = dict(os.environ)
+    env.update({
+        "LD_LIBRARY_PATH": str(FAKEDIR),
+        "LD_PRELOAD": str(LIBVGPU),
+        "VGPU_DEVICE_MEMORY_SHARED_CACHE": str(cache),
+        "VGPU_REAL_HIP_PATH": str(FAKEDIR / "libamdhip64.so"),
+        "VGPU_DEVICE_MEMORY_LIMIT": limit,
+        "VGPU_DEVICE_UUIDS": "GPU-test-0",
+        "VGPU_TASK_PRIORITY": "1",
+    })
+    env.update(extra or {})
+    return env
+
+
+class TestRegionMirror:
+    def test_snapshot_matches_c_writer(self, tmp_path):
+        cache = tmp_path / "r.cache"
+        proc = subprocess.Popen(
+            [str(CONSUMER), "alloc", str(300 * MIB), "sleep", "8000"],
+            env=consumer_env(cache), stdout=subprocess.PIPE, text=True)
+        try:
+            line = json.loads(proc.stdout.readline())
+            assert line["err"] == 0
+            region = SharedRegion(str(cache))
+            assert region.valid
+            snap = region.snapshot()
+            assert snap.limit[0] == 1000 * MIB
+            assert snap.uuids[0] == "GPU-test-0"
+            assert snap.priority == 1
+            assert len(snap.procs) == 1
+            assert snap.procs[0].pid == proc.pid
+            assert snap.device_usage(0) == 300 * MIB
+            region.close()
+        finally:
+            proc.kill()
+            proc.wait()
+
+    def test_layout_consistent(self):
+        layout = region_layout()
+        assert layout["_size"] > layout["procs"]
+        assert layout["_devmem_size"] == 40  # 5 x u64
+
+
+class TestFeedbackBlocking:
+    def test_block_and_unblock_launches(self, tmp_path):
+        """Writing recent_kernel=-1 must stall the consumer's launch hook;
+        restoring it releases the stall (priority preemption path)."""
+        cache = tmp_path / "r.cache"
+        # consumer: create region, wait, then launch storm
+        proc = subprocess.Popen(
+            [str(CONSUMER), "meminfo", "sleep", "1000", "launch", "50", "1"],
+            env=consumer_env(cache), stdout=subprocess.PIPE, text=True)
+        try:
+            json.loads(proc.stdout.readline())  # meminfo -> region exists
+            region = SharedRegion(str(cache))
+            region.set_recent_kernel(-1)  # block before the storm starts
+
+            def unblock():
+                time.sleep(2.5)
+                region.set_recent_kernel(0)
+
+            t = threading.Thread(target=unblock)
+            t.start()
+            out, _ = proc.communicate(timeout=60)
+            t.join()
+            lines = [json.loads(l) for l in out.splitlines()]
+            storm = [l for l in lines if l.get("cmd") == "launch"][0]
+            # storm started at ~1s, unblock at ~2.5s -> >= ~1.2s spent gated
+            assert storm["seconds"] > 1.0, storm
+            assert storm["err"] == 0
+            region.close()
+        finally:
+            if proc.poll() is None:
+                proc.kill()
+                proc.wait()
+
+    def test_feedback_loop_blocks_low_priority(self, tmp_path):
+        """Two containers on one device uuid: high-priority active =>
+        low-priority region gets recent_kernel=-1."""
+        hook = tmp_path / "hook"
+        hi_dir = hook / "containers" / "podA_main"
+        lo_dir = hook / "containers" / "podB_main"
+        hi_dir.mkdir(parents=True)
+        lo_dir.mkdir(parents=True)
+        hi = subprocess.Popen(
+            [str(CONSUMER), "launch", "5", "1", "sleep", "8000"],
+            env=consumer_env(hi_dir / "a.cache", extra={"VGPU_TASK_PRIORITY": "0"}),
+            stdout=subprocess.PIPE, text=True)
+        lo = subprocess.Popen(
+            [str(CONSUMER), "launch", "5", "1", "sleep", "8000"],
+            env=consumer_env(lo_dir / "b.cache", extra={"VGPU_TASK_PRIORITY": "1"}),
+            stdout=subprocess.PIPE, text=True)
+        try:
+            hi.stdout.readline()
+            lo.stdout.readline()
+            pm = PathMonitor(str(hook))
+            pm.scan({"podA", "podB"})
+            assert len(pm.live_regions()) == 2
+            fb = FeedbackLoop(pm)
+            fb.observe_once()
+            lo_region = SharedRegion(str(lo_dir / "b.cache"))
+            hi_region = SharedRegion(str(hi_dir / "a.cache"))
+            assert lo_region.get_recent_kernel() == -1
+            assert hi_region.get_recent_kernel() >= 0
+            # high-priority goes idle -> decay -> low gets unblocked
+            hi_region.set_recent_kernel(0)
+            fb.observe_once()
+            assert lo_region.get_recent_kernel() >= 0
+            lo_region.close()
+            hi_region.close()
+        finally:
+            for p in (hi, lo):
+                p.kill()
+                p.wait()
+
+
+class TestPathMonitor:
+    def _mk_region(self, d, name="x.cache"):
+        d.mkdir(parents=True, exist_ok=True)
+        out = subprocess.run(
+            [str(CONSUMER), "meminfo"], env=consumer_env(d / name),
+            capture_output=True, text=True, timeout=60)
+        assert out.returncode == 0
+
+    def test_gc_after_grace(self, tmp_path):
+        hook = tmp_path / "hook"
+        d = hook / "containers" / "poddead_main"
+        self._mk_region(d)
+        pm = PathMonitor(str(hook))
+        now = time.time()
+        pm.scan(set(), now=now)
+        assert d.exists()  # within grace
+        pm.scan(set(), now=now + GC_GRACE_SECONDS + 1)
+        assert not d.exists()
+
+    def test_live_pod_not_gced(self, tmp_path):
+        hook = tmp_path / "hook"
+        d = hook / "containers" / "podlive_main"
+        self._mk_region(d)
+        pm = PathMonitor(str(hook))
+        now = time.time()
+        pm.scan({"podlive"}, now=now)
+        pm.scan({"podlive"}, now=now + GC_GRACE_SECONDS + 100)
+        assert d.exists()
+
+
+class TestMonitorMetrics:
+    def test_metrics_families(self, tmp_path):
+        hook = tmp_path / "hook"
+        d = hook / "containers" / "podm_main"
+        d.mkdir(parents=True)
+        proc = subprocess.Popen(
+            [str(CONSUMER), "alloc", str(100 * MIB), "sleep", "8000"],
+            env=consumer_env(d / "m.cache"), stdout=subprocess.PIPE, text=True)
+        try:
+            json.loads(proc.stdout.readline())
+            pm = PathMonitor(str(hook))
+            pm.scan({"podm"})
+            text = metrics_text(MonitorCollector(pm, gpus=[])).decode()
+            assert "vGPU_device_memory_usage_in_bytes" in text
+            assert "vGPU_device_memory_limit_in_bytes" in text
+            assert 'poduid="podm"' in text
+            assert ("1.048576e+08" in text) or (str(100 * MIB) in text)
+        finally:
+            proc.kill()
+            proc.wait()
