@@ -30,10 +30,17 @@ _devices: Dict[str, AMDDevices] = {}
 KNOWN_DEVICES: Dict[str, str] = {}
 
 
-def init_devices() -> None:
-    if _devices:
+def init_devices(**config_overrides) -> None:
+    """Initialise the vendor registry.  Keyword overrides (the reference's
+    GlobalFlagSet, devices.go:93-101) set AMDConfig fields, e.g.
+    ``init_devices(resource_name="amd.com/gpu", default_mem=1024)``; a repeat
+    call with no overrides keeps the existing registry."""
+    if _devices and not config_overrides:
         return
-    dev = AMDDevices()
+    from .amd import AMDConfig
+
+    cfg = AMDConfig(**config_overrides) if config_overrides else AMDConfig()
+    dev = AMDDevices(cfg)
     _devices[AMD_DEVICE_TYPE] = dev
     KNOWN_DEVICES[HANDSHAKE_ANNO] = REGISTER_ANNO
 
