@@ -250,3 +250,54 @@ def step(case: BenchCase, model: nn.Module, batch, optimizer=None) -> None:
         if optimizer is not None:
             optimizer.step()
             optimizer.zero_grad(set_to_none=True)
+
+
+# ---------------------------------------------------------------------------
+# CLI: the in-cluster benchmark entry point used by benchmarks/ai-benchmark/*
+# (the reference runs the ai-benchmark binary inside the Job; here the zoo is
+# the suite).  --steps 0 loops forever (density/serving pods).
+# ---------------------------------------------------------------------------
+def main(argv=None):
+    import argparse
+    import json
+    import time
+
+    p = argparse.ArgumentParser("ai-benchmark-zoo")
+    p.add_argument("--cases", default="all", help="comma list or 'all'")
+    p.add_argument("--steps", type=int, default=50,
+                   help="timed steps per case; 0 = run forever")
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--device", default="cuda:0")
+    args = p.parse_args(argv)
+
+    names = list(CASES) if args.cases == "all" else [
+        c.strip() for c in args.cases.split(",") if c.strip()]
+    dev = torch.device(args.device)
+    results = {}
+    for name in names:
+        case = CASES[name]
+        model = build(case, dev)
+        batch = synthetic_batch(case, dev)
+        opt = (torch.optim.SGD(model.parameters(), lr=0.01, momentum=0.9)
+               if case.phase == "training" else None)
+        for _ in range(args.warmup):
+            step(case, model, batch, opt)
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
+        if args.steps == 0:
+            print(f"{name}: serving forever", flush=True)
+            while True:
+                step(case, model, batch, opt)
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            step(case, model, batch, opt)
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        results[name] = case.batch * args.steps / dt
+        print(f"{name}: {results[name]:.2f} samples/s", flush=True)
+    print(json.dumps({"samples_per_s": results}), flush=True)
+
+
+if __name__ == "__main__":
+    main()
