@@ -78,6 +78,20 @@ void *vgpu_real_rsmi_handle(void) {
   return g_real_rsmi;
 }
 
+__thread int vgpu_tls_passthrough = 0;
+
+static void *g_real_hsa = NULL;
+
+void *vgpu_real_hsa(const char *sym) {
+  if (!g_real_hsa) {
+    static const char *const names[] = {"libhsa-runtime64.so.1",
+                                        "libhsa-runtime64.so", NULL};
+    g_real_hsa = open_real(ENV_REAL_HSA, names, "libhsa-runtime64");
+  }
+  if (!g_real_hsa) return NULL;
+  return dlsym(g_real_hsa, sym);
+}
+
 static void do_init(void) {
   if (vgpu_control_disabled()) {
     vgpu_log(VGPU_INFO, "control disabled via %s", ENV_DISABLE_CONTROL);
@@ -113,7 +127,8 @@ static void *g_self_handle = NULL; /* handle apps got from a redirect */
 
 void *dlopen(const char *filename, int flags) {
   if (filename && !tls_no_redirect && !vgpu_control_disabled()) {
-    if (strstr(filename, "libamdhip64") || strstr(filename, "librocm_smi64")) {
+    if (strstr(filename, "libamdhip64") || strstr(filename, "librocm_smi64") ||
+        strstr(filename, "libhsa-runtime64")) {
       const char *self = self_path();
       if (self) {
         vgpu_log(VGPU_INFO, "redirecting dlopen(%s) to %s", filename, self);
@@ -123,6 +138,8 @@ void *dlopen(const char *filename, int flags) {
           /* make sure the real library is resolvable for forwarded misses */
           if (strstr(filename, "libamdhip64"))
             vgpu_real_hip("hipGetDeviceCount");
+          else if (strstr(filename, "libhsa-runtime64"))
+            vgpu_real_hsa("hsa_init");
           else
             vgpu_real_rsmi_handle();
           return h;
@@ -160,6 +177,8 @@ void *dlsym(void *handle, const char *symbol) {
     if (strncmp(symbol, "rsmi_", 5) == 0 || strncmp(symbol, "amdsmi_", 7) == 0) {
       void *h = vgpu_real_rsmi_handle();
       if (h) p = real(h, symbol);
+    } else if (strncmp(symbol, "hsa_", 4) == 0) {
+      p = vgpu_real_hsa(symbol);
     } else {
       p = vgpu_real_hip(symbol);
     }

@@ -145,6 +145,7 @@ static hipError_t alloc_common(void **ptr, size_t size, const char *via,
   if (!vgpu_control_disabled() && vgpu_oom_check(dev, size) != 0)
     return hipErrorOutOfMemory;
   hipError_t e;
+  vgpu_tls_passthrough++; /* the HIP runtime allocates via HSA underneath */
   if (!vgpu_control_disabled() && oversubscribe_mode()) {
     /* managed allocation: XNACK pages beyond-HBM working sets to host DRAM */
     fn_malloc managed = (fn_malloc)vgpu_real_hip("hipMallocManaged");
@@ -152,6 +153,7 @@ static hipError_t alloc_common(void **ptr, size_t size, const char *via,
   } else {
     e = real_fn(ptr, size);
   }
+  vgpu_tls_passthrough--;
   if (e == hipSuccess && !vgpu_control_disabled()) {
     account_alloc(*ptr, size, dev);
     vgpu_log(VGPU_DEBUG, "%s(%zu) dev=%d -> %p", via, size, dev, *ptr);
@@ -170,7 +172,9 @@ hipError_t hipExtMallocWithFlags(void **ptr, size_t size, unsigned int flags) {
   int dev = vgpu_current_device();
   if (!vgpu_control_disabled() && vgpu_oom_check(dev, size) != 0)
     return hipErrorOutOfMemory;
+  vgpu_tls_passthrough++;
   hipError_t e = real_hipExtMallocWithFlags(ptr, size, flags);
+  vgpu_tls_passthrough--;
   if (e == hipSuccess && !vgpu_control_disabled()) account_alloc(*ptr, size, dev);
   return e;
 }
@@ -184,7 +188,9 @@ hipError_t hipMallocManaged(void **ptr, size_t size, unsigned int flags) {
   int dev = vgpu_current_device();
   if (!vgpu_control_disabled() && vgpu_oom_check(dev, size) != 0)
     return hipErrorOutOfMemory;
+  vgpu_tls_passthrough++;
   hipError_t e = real_hipMallocManaged(ptr, size, flags);
+  vgpu_tls_passthrough--;
   if (e == hipSuccess && !vgpu_control_disabled()) account_alloc(*ptr, size, dev);
   return e;
 }
@@ -195,7 +201,9 @@ hipError_t hipMallocAsync(void **ptr, size_t size, void *stream) {
   int dev = vgpu_current_device();
   if (!vgpu_control_disabled() && vgpu_oom_check(dev, size) != 0)
     return hipErrorOutOfMemory;
+  vgpu_tls_passthrough++;
   hipError_t e = real_hipMallocAsync(ptr, size, stream);
+  vgpu_tls_passthrough--;
   if (e == hipSuccess && !vgpu_control_disabled()) account_alloc(*ptr, size, dev);
   return e;
 }
@@ -208,7 +216,9 @@ hipError_t hipMallocFromPoolAsync(void **ptr, size_t size, void *pool,
   int dev = vgpu_current_device();
   if (!vgpu_control_disabled() && vgpu_oom_check(dev, size) != 0)
     return hipErrorOutOfMemory;
+  vgpu_tls_passthrough++;
   hipError_t e = real_hipMallocFromPoolAsync(ptr, size, pool, stream);
+  vgpu_tls_passthrough--;
   if (e == hipSuccess && !vgpu_control_disabled()) account_alloc(*ptr, size, dev);
   return e;
 }
@@ -223,7 +233,9 @@ hipError_t hipMallocPitch(void **ptr, size_t *pitch, size_t width,
   uint64_t est = ((width + 255) & ~255ULL) * height;
   if (!vgpu_control_disabled() && vgpu_oom_check(dev, est) != 0)
     return hipErrorOutOfMemory;
+  vgpu_tls_passthrough++;
   hipError_t e = real_hipMallocPitch(ptr, pitch, width, height);
+  vgpu_tls_passthrough--;
   if (e == hipSuccess && !vgpu_control_disabled())
     account_alloc(*ptr, (uint64_t)(*pitch) * height, dev);
   return e;
@@ -242,7 +254,9 @@ static void account_free(void *ptr) {
 hipError_t hipFree(void *ptr) {
   REAL(fn_free, hipFree);
   vgpu_ensure_initialized();
+  vgpu_tls_passthrough++;
   hipError_t e = real_hipFree(ptr);
+  vgpu_tls_passthrough--;
   if (e == hipSuccess) account_free(ptr);
   return e;
 }
@@ -250,7 +264,9 @@ hipError_t hipFree(void *ptr) {
 hipError_t hipFreeAsync(void *ptr, void *stream) {
   REAL(fn_free_async, hipFreeAsync);
   vgpu_ensure_initialized();
+  vgpu_tls_passthrough++;
   hipError_t e = real_hipFreeAsync(ptr, stream);
+  vgpu_tls_passthrough--;
   if (e == hipSuccess) account_free(ptr);
   return e;
 }

@@ -44,6 +44,7 @@ extern "C" {
 #define ENV_LOG_LEVEL "LIBVGPU_LOG_LEVEL"
 #define ENV_REAL_HIP "VGPU_REAL_HIP_PATH"             /* test hook: fake lib */
 #define ENV_REAL_RSMI "VGPU_REAL_RSMI_PATH"
+#define ENV_REAL_HSA "VGPU_REAL_HSA_PATH"
 #define ENV_DEVICE_UUIDS "VGPU_DEVICE_UUIDS"          /* comma list, monitor correlation */
 
 /* memory accounting split, per device per process (reference ABI:
@@ -111,6 +112,10 @@ void vgpu_limiter_gate(int dev, uint64_t workgroups); /* blocks when throttled *
 /* ---- hook core (hook.c) ---- */
 void *vgpu_real_hip(const char *sym);   /* resolve real libamdhip64 symbol */
 void *vgpu_real_rsmi_handle(void);
+void *vgpu_real_hsa(const char *sym);   /* resolve real libhsa-runtime64 symbol */
+/* >0 while inside one of our own wrappers: lower-layer hooks (hsa.c) must
+ * pass through, not double-count (the HIP runtime allocates via HSA). */
+extern __thread int vgpu_tls_passthrough;
 int vgpu_initialized(void);
 void vgpu_ensure_initialized(void);
 int vgpu_control_disabled(void);

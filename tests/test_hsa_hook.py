@@ -1,0 +1,104 @@
+"""HSA-layer hook tests against the fake libhsa-runtime64.
+
+The interceptor must catch libraries that allocate below HIP —
+hsa_amd_memory_pool_allocate/free — and keep the same shared-region ledger
+(SURVEY.md §7 hard part 1), while NOT double-counting the HIP runtime's own
+internal HSA allocations (vgpu_tls_passthrough guard).
+"""
+import ctypes
+import os
+import subprocess
+from pathlib import Path
+
+import pytest
+
+CSRC = Path(__file__).resolve().parent.parent / "k8s_device_plugin_amd" / "csrc"
+LIBVGPU = CSRC / "libvgpu-hip.so"
+FAKEHSA = CSRC / "fakehsa"
+CONSUMER = CSRC / "test" / "hsa_consumer"
+
+MIB = 1024 * 1024
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    if not (LIBVGPU.exists() and CONSUMER.exists()):
+        subprocess.run(
+            ["make", "libvgpu-hip.so", "fakehsa/libhsa-runtime64.so",
+             "test/hsa_consumer"],
+            cwd=CSRC, check=True, capture_output=True,
+        )
+
+
+def run_consumer(mbs, cache, mem_limit=None, extra_env=None, preload=True):
+    env = dict(os.environ)
+    env["LD_LIBRARY_PATH"] = str(FAKEHSA)
+    if preload:
+        env["LD_PRELOAD"] = str(LIBVGPU)
+    env["VGPU_DEVICE_MEMORY_SHARED_CACHE"] = str(cache)
+    env["VGPU_REAL_HSA_PATH"] = str(FAKEHSA / "libhsa-runtime64.so")
+    # HIP layer unused by this consumer, but the hook resolves the current
+    # device through it; point it at the fake HIP runtime.
+    env["VGPU_REAL_HIP_PATH"] = str(CSRC / "fakehip" / "libamdhip64.so")
+    if mem_limit is not None:
+        env["VGPU_DEVICE_MEMORY_LIMIT"] = mem_limit
+    env.update(extra_env or {})
+    out = subprocess.run(
+        [str(CONSUMER)] + [str(m) for m in mbs],
+        env=env, capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stderr
+    return out.stdout.splitlines()
+
+
+def region_usage(cache):
+    """Read device-0 usage back out of the shared region via the library."""
+    lib = ctypes.CDLL(str(LIBVGPU))
+    os.environ["VGPU_DEVICE_MEMORY_SHARED_CACHE"] = str(cache)
+    lib.vgpu_current_usage.restype = ctypes.c_uint64
+    lib.vgpu_current_usage.argtypes = [ctypes.c_int]
+    return None  # in-process attach not used; tests assert via consumer output
+
+
+class TestHSAQuota:
+    def test_over_quota_alloc_fails(self, tmp_path):
+        lines = run_consumer([600, 600], tmp_path / "r.cache", mem_limit="1000m")
+        assert lines[0] == "ok 0"
+        assert lines[1] == "oom 1"
+
+    def test_under_quota_all_succeed(self, tmp_path):
+        lines = run_consumer([100, 100, 100], tmp_path / "r.cache",
+                             mem_limit="1000m")
+        assert lines[:3] == ["ok 0", "ok 1", "ok 2"]
+
+    def test_free_credits_back(self, tmp_path):
+        # 600 alloc'd then freed (consumer frees even indices) -> region
+        # usage returns to the odd allocations only; a second consumer can
+        # then allocate 600 against the same region file.
+        cache = tmp_path / "r.cache"
+        lines = run_consumer([600, 300], cache, mem_limit="1000m")
+        assert lines[:2] == ["ok 0", "ok 1"]
+        # ptr 0 (600M) freed at exit of consumer 1; ptr 1 (300M) freed too
+        # when the process died (region GC: proc slot cleanup on next attach)
+        lines = run_consumer([900], cache, mem_limit="1000m")
+        assert lines[0] == "ok 0"
+
+    def test_no_preload_uncapped(self, tmp_path):
+        lines = run_consumer([600, 600], tmp_path / "r.cache",
+                             mem_limit="1000m", preload=False)
+        assert lines[:2] == ["ok 0", "ok 1"]
+
+    def test_disable_control_passthrough(self, tmp_path):
+        lines = run_consumer([600, 600], tmp_path / "r.cache",
+                             mem_limit="1000m",
+                             extra_env={"VGPU_DISABLE_CONTROL": "1"})
+        assert lines[:2] == ["ok 0", "ok 1"]
+
+
+def test_exports_present():
+    """The .so exports the HSA hook symbols (so PLT interposition works)."""
+    out = subprocess.run(["nm", "-D", str(LIBVGPU)], capture_output=True,
+                         text=True, check=True)
+    for sym in ["hsa_amd_memory_pool_allocate", "hsa_amd_memory_pool_free",
+                "hsa_memory_allocate", "hsa_memory_free"]:
+        assert sym in out.stdout
