@@ -60,6 +60,8 @@ def main(argv=None):
     p = argparse.ArgumentParser("vgpu-monitor")
     p.add_argument("--hook-path", default=os.environ.get("HOOK_PATH", "/usr/local/vgpu"))
     p.add_argument("--metrics-port", type=int, default=9394)
+    p.add_argument("--grpc-bind", default="0.0.0.0:9395",
+                   help="NodeVGPUInfo gRPC bind ('' disables)")
     p.add_argument("--interval", type=float, default=5.0)
     p.add_argument("--soft-cores", action="store_true",
                    default=os.environ.get("VGPU_MONITOR_SOFT_CORES", "") == "1")
@@ -74,7 +76,13 @@ def main(argv=None):
     feedback = FeedbackLoop(pathmon, soft_cores=args.soft_cores)
     collector = MonitorCollector(pathmon, gpus)
     serve_metrics(collector, args.metrics_port)
-    log.info("monitor up: %d GPUs, metrics :%d", len(gpus), args.metrics_port)
+    if args.grpc_bind:
+        from . import noderpc
+
+        noderpc.serve(pathmon, node_name=os.environ.get("NODE_NAME", ""),
+                      bind=args.grpc_bind)
+    log.info("monitor up: %d GPUs, metrics :%d, grpc %s", len(gpus),
+             args.metrics_port, args.grpc_bind or "off")
 
     client: Optional[KubeClient] = None
     try:

@@ -192,3 +192,47 @@ class TestMonitorMetrics:
         finally:
             proc.kill()
             proc.wait()
+
+
+class TestNodeRPC:
+    def test_get_node_vgpu_over_grpc(self, tmp_path):
+        """NodeVGPUInfo serves live shared-region state (reference
+        noderpc.proto:25-61, but actually implemented here)."""
+        import grpc
+
+        from k8s_device_plugin_amd.monitor import noderpc
+
+        hook = tmp_path / "hook"
+        ctr_dir = hook / "containers" / "pod-rpc-1_main"
+        ctr_dir.mkdir(parents=True)
+        cache = ctr_dir / "r.cache"
+        proc = subprocess.Popen(
+            [str(CONSUMER), "alloc", str(200 * MIB), "sleep", "8000"],
+            env=consumer_env(cache), stdout=subprocess.PIPE, text=True)
+        server = None
+        try:
+            line = json.loads(proc.stdout.readline())
+            assert line["err"] == 0
+            pathmon = PathMonitor(str(hook))
+            pathmon.scan({"pod-rpc-1"})
+            server, port = noderpc.serve(pathmon, node_name="n1",
+                                         bind="127.0.0.1:0")
+            with grpc.insecure_channel(f"127.0.0.1:{port}") as ch:
+                client = noderpc.NodeVGPUClient(ch)
+                reply = client.get_node_vgpu()
+                assert reply.nodeid == "n1"
+                assert len(reply.nodevgpuinfo) == 1
+                pu = reply.nodevgpuinfo[0]
+                assert pu.poduuid == "pod-rpc-1"
+                assert pu.container == "main"
+                sr = pu.podvgpuinfo
+                assert sr.limit[0] == 1000 * MIB
+                assert sum(p.used[0] for p in sr.procs) == 200 * MIB
+                # filter by pod uid
+                reply = client.get_node_vgpu(ctruuid="nope")
+                assert len(reply.nodevgpuinfo) == 0
+        finally:
+            if server:
+                server.stop(0)
+            proc.kill()
+            proc.wait()
