@@ -34,6 +34,11 @@ class PluginConfig:
     config_file: str = "/config/config.json"
     register_interval_s: float = 30.0
     health_interval_s: float = 5.0
+    # "envvar" (ROCR_VISIBLE_DEVICES + DeviceSpecs) or "cdi-annotations"
+    # (additionally name CDI devices in the Allocate response annotations;
+    # reference --device-list-strategy, main.go:61-70)
+    device_list_strategy: str = "envvar"
+    cdi_spec_dir: str = "/var/run/cdi"
 
 
 def parse_args(argv: Optional[List[str]] = None) -> PluginConfig:
@@ -49,6 +54,9 @@ def parse_args(argv: Optional[List[str]] = None) -> PluginConfig:
     p.add_argument("--plugin-socket-dir", default=c.plugin_socket_dir)
     p.add_argument("--kubelet-socket", default=c.kubelet_socket)
     p.add_argument("--config-file", default=c.config_file)
+    p.add_argument("--device-list-strategy", default=c.device_list_strategy,
+                   choices=["envvar", "cdi-annotations"])
+    p.add_argument("--cdi-spec-dir", default=c.cdi_spec_dir)
     a = p.parse_args(argv)
     cfg = PluginConfig(
         node_name=a.node_name,
@@ -61,6 +69,8 @@ def parse_args(argv: Optional[List[str]] = None) -> PluginConfig:
         plugin_socket_dir=a.plugin_socket_dir,
         kubelet_socket=a.kubelet_socket,
         config_file=a.config_file,
+        device_list_strategy=a.device_list_strategy,
+        cdi_spec_dir=a.cdi_spec_dir,
     )
     return apply_node_config(cfg)
 

@@ -101,6 +101,13 @@ class PluginDaemon:
         ('stop' | 'reload' | 'kubelet-restart')."""
         cfg = self.cfg
         rm = build_resource_manager(cfg)
+        try:
+            from . import cdi as cdimod
+
+            if rm.gpus:
+                cdimod.write_spec(rm.gpus, cfg.cdi_spec_dir, cfg.hook_path)
+        except OSError as e:
+            log.warning("CDI spec write failed: %s", e)
         plugin = VGPUDevicePlugin(cfg, rm, self.client)
         plugin.serve()
         try:

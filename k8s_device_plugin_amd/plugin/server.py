@@ -192,6 +192,11 @@ class VGPUDevicePlugin:
             resp.mounts.add(container_path="/etc/ld.so.preload",
                             host_path=f"{cfg.hook_path}/vgpu/ld.so.preload",
                             read_only=True)
+        if cfg.device_list_strategy == "cdi-annotations":
+            from . import cdi as cdimod
+
+            for k, v in cdimod.annotations(visible).items():
+                resp.annotations[k] = v
         return resp
 
     def release_pod(self, pod_uid: str) -> None:
