@@ -270,3 +270,83 @@ hipError_t hipFreeAsync(void *ptr, void *stream) {
   if (e == hipSuccess) account_free(ptr);
   return e;
 }
+
+/* ---- array allocations (the reference accounts cuArray{,3D}Create /
+ * cuMipmappedArrayCreate the same way, SURVEY.md §2.6 "Memory cap") ---- */
+typedef struct { size_t width, height, depth; } vgpu_extent_t;
+typedef struct { void *ptr; size_t pitch, xsize, ysize; } vgpu_pitched_ptr_t;
+typedef struct { int x, y, z, w; int f; } vgpu_chan_desc_t;
+
+static uint64_t chan_bytes(const vgpu_chan_desc_t *d) {
+  if (!d) return 4;
+  int bits = d->x + d->y + d->z + d->w;
+  return bits > 0 ? (uint64_t)(bits + 7) / 8 : 4;
+}
+
+hipError_t hipMalloc3D(vgpu_pitched_ptr_t *p, vgpu_extent_t extent) {
+  typedef hipError_t (*fn)(vgpu_pitched_ptr_t *, vgpu_extent_t);
+  REAL(fn, hipMalloc3D);
+  vgpu_ensure_initialized();
+  int dev = vgpu_current_device();
+  uint64_t est = ((extent.width + 255) & ~255ULL) * extent.height *
+                 (extent.depth ? extent.depth : 1);
+  if (!vgpu_control_disabled() && vgpu_oom_check(dev, est) != 0)
+    return hipErrorOutOfMemory;
+  vgpu_tls_passthrough++;
+  hipError_t e = real_hipMalloc3D(p, extent);
+  vgpu_tls_passthrough--;
+  if (e == hipSuccess && !vgpu_control_disabled())
+    account_alloc(p->ptr,
+                  p->pitch * extent.height * (extent.depth ? extent.depth : 1),
+                  dev);
+  return e;
+}
+
+hipError_t hipMallocArray(void **array, const vgpu_chan_desc_t *desc,
+                          size_t width, size_t height, unsigned int flags) {
+  typedef hipError_t (*fn)(void **, const vgpu_chan_desc_t *, size_t, size_t,
+                           unsigned int);
+  REAL(fn, hipMallocArray);
+  vgpu_ensure_initialized();
+  int dev = vgpu_current_device();
+  uint64_t bytes = chan_bytes(desc) * width * (height ? height : 1);
+  if (!vgpu_control_disabled() && vgpu_oom_check(dev, bytes) != 0)
+    return hipErrorOutOfMemory;
+  vgpu_tls_passthrough++;
+  hipError_t e = real_hipMallocArray(array, desc, width, height, flags);
+  vgpu_tls_passthrough--;
+  if (e == hipSuccess && !vgpu_control_disabled())
+    account_alloc(*array, bytes, dev);
+  return e;
+}
+
+hipError_t hipMalloc3DArray(void **array, const vgpu_chan_desc_t *desc,
+                            vgpu_extent_t extent, unsigned int flags) {
+  typedef hipError_t (*fn)(void **, const vgpu_chan_desc_t *, vgpu_extent_t,
+                           unsigned int);
+  REAL(fn, hipMalloc3DArray);
+  vgpu_ensure_initialized();
+  int dev = vgpu_current_device();
+  uint64_t bytes = chan_bytes(desc) * extent.width *
+                   (extent.height ? extent.height : 1) *
+                   (extent.depth ? extent.depth : 1);
+  if (!vgpu_control_disabled() && vgpu_oom_check(dev, bytes) != 0)
+    return hipErrorOutOfMemory;
+  vgpu_tls_passthrough++;
+  hipError_t e = real_hipMalloc3DArray(array, desc, extent, flags);
+  vgpu_tls_passthrough--;
+  if (e == hipSuccess && !vgpu_control_disabled())
+    account_alloc(*array, bytes, dev);
+  return e;
+}
+
+hipError_t hipFreeArray(void *array) {
+  typedef hipError_t (*fn)(void *);
+  REAL(fn, hipFreeArray);
+  vgpu_ensure_initialized();
+  vgpu_tls_passthrough++;
+  hipError_t e = real_hipFreeArray(array);
+  vgpu_tls_passthrough--;
+  if (e == hipSuccess) account_free(array);
+  return e;
+}
