@@ -27,6 +27,7 @@ log = logging.getLogger(__name__)
 
 KFD_TOPOLOGY = "/sys/class/kfd/kfd/topology"
 PCI_DEVICES = "/sys/bus/pci/devices"
+DRM_CLASS = "/sys/class/drm"
 KFD_DEV = "/dev/kfd"
 
 
@@ -43,6 +44,10 @@ class PhysicalGPU:
     drm_render_minor: int   # /dev/dri/renderD<minor>
     gfx_target: str         # e.g. "gfx950"
     io_links: Dict[int, int] = field(default_factory=dict)  # node_to -> type
+    # MI300+/MI355X compute/memory partitioning (the MIG analog: in CPX each
+    # XCD is its own KFD node; amdgpu exposes the mode per drm card)
+    compute_partition: str = "SPX"   # SPX / DPX / QPX / CPX
+    memory_partition: str = "NPS1"   # NPS1 / NPS2 / NPS4
 
     @property
     def drm_card(self) -> int:
@@ -81,8 +86,17 @@ def _read_int(path: str, default: int = -1) -> int:
         return default
 
 
+def _read_str(path: str, default: str = "") -> str:
+    try:
+        with open(path) as f:
+            return f.read().strip()
+    except OSError:
+        return default
+
+
 def enumerate_gpus(
-    topology_root: str = KFD_TOPOLOGY, pci_root: str = PCI_DEVICES
+    topology_root: str = KFD_TOPOLOGY, pci_root: str = PCI_DEVICES,
+    drm_root: str = DRM_CLASS,
 ) -> List[PhysicalGPU]:
     gpus: List[PhysicalGPU] = []
     nodes_dir = os.path.join(topology_root, "nodes")
@@ -125,6 +139,12 @@ def enumerate_gpus(
             lprops = _read_properties(link)
             if "node_to" in lprops:
                 io_links[lprops["node_to"]] = lprops.get("type", 0)
+        render_minor = props.get("drm_render_minor", 128 + len(gpus))
+        card_dev = os.path.join(drm_root, f"card{render_minor - 128}", "device")
+        compute_part = _read_str(
+            os.path.join(card_dev, "current_compute_partition"), "SPX") or "SPX"
+        memory_part = _read_str(
+            os.path.join(card_dev, "current_memory_partition"), "NPS1") or "NPS1"
         gpus.append(
             PhysicalGPU(
                 index=len(gpus),
@@ -135,9 +155,11 @@ def enumerate_gpus(
                 mem_bytes=mem_bytes,
                 numa_node=numa,
                 pci_bdf=bdf,
-                drm_render_minor=props.get("drm_render_minor", 128 + len(gpus)),
+                drm_render_minor=render_minor,
                 gfx_target=gfx,
                 io_links=io_links,
+                compute_partition=compute_part,
+                memory_partition=memory_part,
             )
         )
     return gpus

@@ -70,12 +70,19 @@ class ResourceManager:
         (reference getApiDevices, register.go:96-162)."""
         out = []
         for idx, g in enumerate(self.gpus):
+            # partitioned cards advertise a distinct type so pods can pin or
+            # avoid them via use/nouse-gputype (the reference's MIG
+            # mixed-strategy analog, rm/device_map.go:95-118): in CPX mode
+            # each XCD is its own KFD node with cu_count 32 and 1/8 the HBM
+            part = getattr(g, "compute_partition", "SPX")
+            dtype = self.device_type if part in ("", "SPX") \
+                else f"{self.device_type}-{part}"
             out.append(DeviceInfo(
                 id=g.uuid,
                 count=self.split_count,
                 devmem=int(g.mem_bytes / MIB * self.memory_scaling),
                 devcore=int(100 * self.cores_scaling),
-                type=self.device_type,
+                type=dtype,
                 numa=g.numa_node,
                 health=self.health.get(g.uuid, True),
                 index=idx,

@@ -347,3 +347,47 @@ class TestGRPCPath:
         assert envs["VGPU_OVERSUBSCRIBE"] == "true"
         assert envs["HSA_XNACK"] == "1"
         assert envs["VGPU_DEVICE_MEMORY_LIMIT_0"] == "409600m"
+
+
+class TestComputePartition:
+    """MI355X SPX/CPX partition modes (the MIG analog: amdgpu exposes
+    current_compute_partition per drm card; in CPX each XCD is its own KFD
+    node).  Partitioned cards advertise a distinct device type so gputype
+    white/blacklists can target them."""
+
+    def _drm_tree(self, root: Path, n, mode="CPX", mem_mode="NPS2"):
+        drm = root / "drm"
+        for i in range(n):
+            d = drm / f"card{i}" / "device"
+            d.mkdir(parents=True)
+            (d / "current_compute_partition").write_text(mode + "\n")
+            (d / "current_memory_partition").write_text(mem_mode + "\n")
+        return drm
+
+    def test_partition_mode_read(self, tmp_path):
+        topo, pci = make_kfd_tree(tmp_path, n_gpus=2)
+        drm = self._drm_tree(tmp_path, 2, "CPX", "NPS2")
+        gpus = enumerate_gpus(str(topo), str(pci), str(drm))
+        assert all(g.compute_partition == "CPX" for g in gpus)
+        assert all(g.memory_partition == "NPS2" for g in gpus)
+
+    def test_default_spx_when_absent(self, tmp_path):
+        topo, pci = make_kfd_tree(tmp_path, n_gpus=1)
+        gpus = enumerate_gpus(str(topo), str(pci), str(tmp_path / "nodrm"))
+        assert gpus[0].compute_partition == "SPX"
+        assert gpus[0].memory_partition == "NPS1"
+
+    def test_partitioned_type_in_registration(self, tmp_path):
+        topo, pci = make_kfd_tree(tmp_path, n_gpus=2)
+        drm = self._drm_tree(tmp_path, 2, "CPX")
+        rm = ResourceManager(enumerate_gpus(str(topo), str(pci), str(drm)),
+                             split_count=4)
+        devs = rm.api_devices()
+        assert all(d.type == "AMD-Instinct-MI355X-CPX" for d in devs)
+
+    def test_spx_type_unchanged(self, tmp_path):
+        topo, pci = make_kfd_tree(tmp_path, n_gpus=1)
+        rm = ResourceManager(enumerate_gpus(str(topo), str(pci),
+                                            str(tmp_path / "nodrm")),
+                             split_count=4)
+        assert rm.api_devices()[0].type == "AMD-Instinct-MI355X"
