@@ -109,6 +109,16 @@ class CoreMaskAllocator:
         self.used[device_uuid] = busy | mask
         return mask
 
+    def adopt(self, device_uuid: str, mask: int) -> bool:
+        """Re-register a mask handed out before a plugin restart
+        (reconciliation, the DCU RefreshContainerDevices analog,
+        dcu/server.go:274-316).  False if it conflicts with live state."""
+        busy = self.used.get(device_uuid, 0)
+        if busy & mask:
+            return False
+        self.used[device_uuid] = busy | mask
+        return True
+
     def free(self, device_uuid: str, mask: int) -> None:
         self.used[device_uuid] = self.used.get(device_uuid, 0) & ~mask
 

@@ -121,7 +121,17 @@ class PluginDaemon:
                                interval_s=cfg.health_interval_s)
         health.start()
         watch = _SocketWatch(cfg.kubelet_socket)
+
+        def reconcile_once():
+            try:
+                live = {p.uid for p in self.client.list_pods()}
+                plugin.reconcile(live)
+            except Exception as e:
+                log.warning("reconcile failed: %s", e)
+
+        reconcile_once()  # adopt pre-restart allocations before new Allocates
         reason = "stop"
+        ticks = 0
         try:
             while not self._stop.is_set():
                 if self._hup.is_set():
@@ -132,6 +142,9 @@ class PluginDaemon:
                     log.info("kubelet socket re-created; restarting plugin")
                     reason = "kubelet-restart"
                     break
+                ticks += 1
+                if ticks % 30 == 0:
+                    reconcile_once()
                 time.sleep(1.0)
         finally:
             health.stop()

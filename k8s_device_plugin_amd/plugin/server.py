@@ -179,6 +179,24 @@ class VGPUDevicePlugin:
             os.makedirs(cache_dir, mode=0o777, exist_ok=True)
             os.chmod(cache_dir, 0o777)
             os.makedirs("/tmp/vgpulock", mode=0o777, exist_ok=True)
+            # allocation record for restart reconciliation (the DCU plugin
+            # persists vdev conf files the same way, dcu/server.go:415-465)
+            import json as _json
+
+            with open(os.path.join(cache_dir, "vgpu.json"), "w") as f:
+                _json.dump({
+                    "pod_uid": pod.uid,
+                    "container": ctr.name,
+                    "devices": [
+                        {"uuid": d.uuid, "usedmem": d.usedmem,
+                         "usedcores": d.usedcores,
+                         "cu_mask": f"{m:x}" if m else ""}
+                        for d, m in zip(
+                            devreq,
+                            [next((mm for ii, mm in mask_assignments if ii == i),
+                                  0) for i in range(len(devreq))])
+                    ],
+                }, f)
         except OSError as e:
             log.warning("cannot create hook dirs: %s", e)
         resp.mounts.add(
@@ -198,6 +216,48 @@ class VGPUDevicePlugin:
             for k, v in cdimod.annotations(visible).items():
                 resp.annotations[k] = v
         return resp
+
+    def reconcile(self, live_pod_uids) -> None:
+        """Rebuild CU-mask allocator state from the per-container records
+        after a plugin restart, and GC records of dead pods (reference DCU
+        RefreshContainerDevices, dcu/server.go:274-316)."""
+        import json as _json
+        import shutil
+
+        root = f"{self.cfg.hook_path}/vgpu/containers"
+        if not os.path.isdir(root):
+            return
+        for name in os.listdir(root):
+            d = os.path.join(root, name)
+            rec_path = os.path.join(d, "vgpu.json")
+            pod_uid, _, _ctr = name.partition("_")
+            if pod_uid not in live_pod_uids:
+                if pod_uid in self.pod_masks:
+                    self.release_pod(pod_uid)
+                try:
+                    shutil.rmtree(d)
+                    log.info("reconcile: removed orphan %s", name)
+                except OSError as e:
+                    log.warning("reconcile: cannot remove %s: %s", d, e)
+                continue
+            if pod_uid in self.pod_masks or not os.path.isfile(rec_path):
+                continue
+            try:
+                with open(rec_path) as f:
+                    rec = _json.load(f)
+            except (OSError, ValueError) as e:
+                log.warning("reconcile: bad record %s: %s", rec_path, e)
+                continue
+            for dev in rec.get("devices", []):
+                mask = int(dev.get("cu_mask") or "0", 16)
+                if not mask:
+                    continue
+                if self.cumask.adopt(dev["uuid"], mask):
+                    self.pod_masks.setdefault(rec["pod_uid"], []).append(
+                        (dev["uuid"], mask))
+                else:
+                    log.warning("reconcile: mask conflict for pod %s dev %s",
+                                rec["pod_uid"], dev["uuid"])
 
     def release_pod(self, pod_uid: str) -> None:
         for uuid, mask in self.pod_masks.pop(pod_uid, []):

@@ -3,6 +3,7 @@ allocator, fake-device fan-out, and the full kubelet gRPC path against a
 stub kubelet socket — BASELINE config 1, the test the reference never had
 (SURVEY.md §4).
 """
+import json
 import os
 import threading
 import time
@@ -391,3 +392,64 @@ class TestComputePartition:
                                             str(tmp_path / "nodrm")),
                              split_count=4)
         assert rm.api_devices()[0].type == "AMD-Instinct-MI355X"
+
+
+class TestReconciliation:
+    """Restart reconciliation: allocation records persist under
+    <hook>/vgpu/containers/<uid>_<ctr>/vgpu.json and a fresh plugin adopts
+    the CU masks (reference DCU RefreshContainerDevices,
+    dcu/server.go:274-316)."""
+
+    def _allocate(self, plugin, client, rm):
+        uuid = rm.gpus[0].uuid
+        devs = [[ContainerDevice(uuid=uuid, type="AMD", usedmem=73728,
+                                 usedcores=25)]]
+        pod = PodInfo(
+            name="p1", uid="uid-p1",
+            containers=[ContainerSpec(name="main", limits={"amd.com/gpu": 1})],
+            annotations={
+                BIND_TIME_ANNO: "123",
+                BIND_PHASE_ANNO: BIND_PHASE_ALLOCATING,
+                ASSIGNED_NODE_ANNO: "node1",
+                IN_REQUEST_DEVICES["AMD"]: encode_pod_single_device(devs),
+            },
+        )
+        client.add_pod(pod)
+        client.patch_node_annotations(
+            "node1", {NODE_LOCK_ANNO: "2026-01-01T00:00:00Z"})
+        with grpc.insecure_channel(f"unix://{plugin.socket_path}") as ch:
+            stub = dp.DevicePluginClient(ch)
+            req = dp.AllocateRequest()
+            req.container_requests.add(devicesIDs=[f"{uuid}-0"])
+            stub.Allocate(req)
+        return uuid
+
+    def test_record_written_and_adopted_after_restart(self, plugin_env, tmp_path):
+        plugin, kubelet, client, rm, cfg = plugin_env
+        uuid = self._allocate(plugin, client, rm)
+        rec = Path(cfg.hook_path) / "vgpu" / "containers" / "uid-p1_main" / "vgpu.json"
+        assert rec.is_file()
+        data = json.loads(rec.read_text())
+        assert data["devices"][0]["uuid"] == uuid
+        mask = int(data["devices"][0]["cu_mask"], 16)
+        assert bin(mask).count("1") == 64  # 25% of 256
+
+        # "restart": a fresh plugin instance over the same hook dir
+        plugin2 = VGPUDevicePlugin(cfg, rm, client)
+        assert plugin2.cumask.used_count(uuid) == 0
+        plugin2.reconcile({"uid-p1"})
+        assert plugin2.cumask.used_count(uuid) == 64
+        assert ("uuid" and uuid) and plugin2.pod_masks["uid-p1"] == [(uuid, mask)]
+        # idempotent
+        plugin2.reconcile({"uid-p1"})
+        assert plugin2.cumask.used_count(uuid) == 64
+
+    def test_orphan_record_removed_and_mask_freed(self, plugin_env):
+        plugin, kubelet, client, rm, cfg = plugin_env
+        uuid = self._allocate(plugin, client, rm)
+        d = Path(cfg.hook_path) / "vgpu" / "containers" / "uid-p1_main"
+        assert d.is_dir()
+        assert plugin.cumask.used_count(uuid) == 64
+        plugin.reconcile(set())  # pod gone
+        assert not d.exists()
+        assert plugin.cumask.used_count(uuid) == 0
