@@ -37,12 +37,30 @@ typedef struct {
   size_t totalGlobalMem;
 } prop_r0000_prefix_t;
 
+/* The current device is consulted on EVERY launch (limiter gate) and every
+ * allocation; calling the real hipGetDevice each time costs a runtime API
+ * round-trip per kernel.  hipSetDevice is interposed below to keep a TLS
+ * shadow, so the hot path is one TLS read. */
+static __thread int tls_current_dev = -1;
+
+hipError_t hipSetDevice(int device) {
+  typedef hipError_t (*fn)(int);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipSetDevice");
+  if (!real) return hipErrorInvalidValue;
+  hipError_t e = real(device);
+  if (e == hipSuccess) tls_current_dev = device;
+  return e;
+}
+
 int vgpu_current_device(void) {
+  if (tls_current_dev >= 0) return tls_current_dev;
   typedef hipError_t (*fn)(int *);
   static fn real = NULL;
   if (!real) real = (fn)vgpu_real_hip("hipGetDevice");
   int dev = 0;
   if (real) real(&dev);
+  tls_current_dev = dev;
   return dev;
 }
 

@@ -245,6 +245,15 @@ void vgpu_limiter_init(void) {
   pthread_mutex_unlock(&g_watch_mu);
 }
 
+static int core_policy_disabled(void) {
+  static int cached = -1; /* getenv once, not per launch */
+  if (cached < 0) {
+    const char *policy = getenv(ENV_CORE_POLICY);
+    cached = (policy && strcasecmp(policy, "disable") == 0) ? 1 : 0;
+  }
+  return cached;
+}
+
 void vgpu_limiter_gate(int dev, uint64_t workgroups) {
   vgpu_region_t *r = vgpu_region_get();
   if (!r || dev < 0 || dev >= VGPU_MAX_DEVICES) return;
@@ -263,8 +272,7 @@ void vgpu_limiter_gate(int dev, uint64_t workgroups) {
 
   uint64_t lim = r->sm_limit[dev];
   if (lim == 0 || lim >= 100) return;
-  const char *policy = getenv(ENV_CORE_POLICY);
-  if (policy && strcasecmp(policy, "disable") == 0) return;
+  if (core_policy_disabled()) return;
   if (__atomic_load_n(&r->utilization_switch, __ATOMIC_RELAXED) == 0)
     return; /* monitor says: uncontended, free-run */
   if (!g_watcher_started) vgpu_limiter_init();
