@@ -1,0 +1,136 @@
+#!/usr/bin/env python3
+"""Density benchmark: N co-located "pods" on one MI355X (BASELINE config 3).
+
+Simulates the published density claim (10 inference pods per GPU,
+reference README.md:40) without a cluster: N worker processes, each under
+LD_PRELOAD libvgpu-hip.so with
+
+  - HBM quota        = quota_pct% of the card (default 100/N, like
+                       gpumem-percentage),
+  - CU soft limit    = 100/N percent (VGPU_DEVICE_CU_LIMIT),
+  - its own shared region (one per "container", as the plugin injects),
+
+each running ResNet-V2-50 inference (b50 @346^2, the ai-benchmark case)
+for --seconds wall-clock.  Reports per-worker and aggregate samples/s plus
+the fairness spread (max/min) — the number the reference only publishes as
+a chart.
+
+Usage: python benchmarks/density_bench.py --pods 10 --seconds 30
+"""
+import argparse
+import json
+import os
+import subprocess
+import sys
+import tempfile
+import time
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+LIBVGPU = REPO / "k8s_device_plugin_amd" / "csrc" / "libvgpu-hip.so"
+
+
+def worker(args):
+    import torch
+
+    sys.path.insert(0, str(REPO))
+    from k8s_device_plugin_amd.models import zoo
+
+    case = zoo.CASES[args.case]
+    dev = torch.device("cuda", 0)
+    model = zoo.build(case, dev)
+    batch = zoo.synthetic_batch(case, dev)
+    for _ in range(3):
+        zoo.step(case, model, batch, None)
+    torch.cuda.synchronize()
+    print("READY", flush=True)
+    assert sys.stdin.readline().strip() == "GO"
+    t0 = time.perf_counter()
+    steps = 0
+    while time.perf_counter() - t0 < args.seconds:
+        zoo.step(case, model, batch, None)
+        steps += 1
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(json.dumps({"samples_per_s": case.batch * steps / dt,
+                      "steps": steps}), flush=True)
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--pods", type=int, default=10)
+    p.add_argument("--seconds", type=float, default=30.0)
+    p.add_argument("--case", default="resnet50_inf")
+    p.add_argument("--quota-pct", type=int, default=0,
+                   help="HBM percent per pod (default 100/pods)")
+    p.add_argument("--cu-pct", type=int, default=0,
+                   help="CU percent per pod (default 100/pods)")
+    p.add_argument("--worker", action="store_true", help=argparse.SUPPRESS)
+    args = p.parse_args()
+    if args.worker:
+        worker(args)
+        return
+
+    import torch
+
+    total_mem = torch.cuda.get_device_properties(0).total_memory \
+        if torch.cuda.is_available() else 288 << 30
+    quota_pct = args.quota_pct or max(1, 100 // args.pods)
+    cu_pct = args.cu_pct or max(1, 100 // args.pods)
+    quota_mib = total_mem * quota_pct // 100 // (1 << 20)
+
+    procs = []
+    caches = []
+    for i in range(args.pods):
+        cache = tempfile.NamedTemporaryFile(prefix=f"density-{i}-",
+                                            suffix=".cache", delete=False)
+        cache.close()
+        caches.append(cache.name)
+        env = dict(os.environ)
+        env.update({
+            "LD_PRELOAD": str(LIBVGPU),
+            "VGPU_DEVICE_MEMORY_LIMIT": f"{quota_mib}m",
+            "VGPU_DEVICE_CU_LIMIT": str(cu_pct),
+            "VGPU_DEVICE_MEMORY_SHARED_CACHE": cache.name,
+            "VGPU_DEVICE_UUIDS": f"GPU-density-{i}",
+        })
+        procs.append(subprocess.Popen(
+            [sys.executable, __file__, "--worker", "--case", args.case,
+             "--seconds", str(args.seconds)],
+            env=env, stdin=subprocess.PIPE, stdout=subprocess.PIPE, text=True))
+
+    # barrier: wait until every pod is warm, then release together
+    for pr in procs:
+        assert pr.stdout.readline().strip() == "READY"
+    for pr in procs:
+        pr.stdin.write("GO\n")
+        pr.stdin.flush()
+
+    results = []
+    for pr in procs:
+        line = ""
+        for line in pr.stdout:
+            if line.startswith("{"):
+                break
+        pr.wait(timeout=120)
+        results.append(json.loads(line))
+    for c in caches:
+        os.unlink(c)
+
+    rates = [r["samples_per_s"] for r in results]
+    print(json.dumps({
+        "metric": "co-located pods/GPU at equal HBM+CU quota",
+        "value": args.pods,
+        "unit": "pods",
+        "aggregate_samples_per_s": round(sum(rates), 2),
+        "per_pod_samples_per_s": [round(r, 2) for r in rates],
+        "fairness_max_over_min": round(max(rates) / max(min(rates), 1e-9), 3),
+        "quota_pct": quota_pct,
+        "cu_pct": cu_pct,
+        "case": args.case,
+        "seconds": args.seconds,
+    }))
+
+
+if __name__ == "__main__":
+    main()
