@@ -65,7 +65,7 @@ class VGPUDevicePlugin:
     # ---- gRPC servicer methods -----------------------------------------
     def GetDevicePluginOptions(self, request, context):
         return dp.DevicePluginOptions(pre_start_required=False,
-                                      get_preferred_allocation_available=False)
+                                      get_preferred_allocation_available=True)
 
     def ListAndWatch(self, request, context):
         while not self._stop.is_set():
@@ -88,8 +88,57 @@ class VGPUDevicePlugin:
         return dp.PreStartContainerResponse()
 
     def GetPreferredAllocation(self, request, context):
-        # parity with the reference: advertised unavailable (server.go:270-285)
-        return dp.PreferredAllocationResponse()
+        """xGMI-aligned preferred allocation.
+
+        The reference has aligned (NVLink) / distributed allocators but
+        leaves the server verb commented out (rm/allocate.go:44-121,
+        server.go:270-285); here it is live: fractional requests pack onto
+        the fewest physical GPUs, multi-GPU requests extend along maximal
+        xGMI connectivity (parallel/topology.py)."""
+        from ..parallel.topology import GPUTopology
+
+        topo = GPUTopology.from_gpus(self.rm.gpus) if self.rm.gpus else None
+        uuid_to_idx = {g.uuid: i for i, g in enumerate(self.rm.gpus)}
+        resp = dp.PreferredAllocationResponse()
+        for creq in request.container_requests:
+            chosen = list(creq.must_include_deviceIDs)
+            by_uuid: Dict[str, List[str]] = {}
+            for fid in creq.available_deviceIDs:
+                if fid in chosen:
+                    continue
+                by_uuid.setdefault(self.rm.uuid_of_fake(fid), []).append(fid)
+            need = creq.allocation_size - len(chosen)
+            chosen_uuids = {self.rm.uuid_of_fake(f) for f in chosen}
+            # 1. pack: drain fakes of GPUs already in the set
+            for u in sorted(chosen_uuids):
+                while need > 0 and by_uuid.get(u):
+                    chosen.append(by_uuid[u].pop(0))
+                    need -= 1
+            # 2. extend: next GPU = max xGMI edges to the current set
+            while need > 0 and by_uuid:
+                cands = [u for u, f in by_uuid.items() if f and u in uuid_to_idx]
+                if not cands:
+                    break
+                if topo is not None and chosen_uuids:
+                    cur = [uuid_to_idx[u] for u in chosen_uuids
+                           if u in uuid_to_idx]
+
+                    def edges(u):
+                        return topo.xgmi_degree(cur + [uuid_to_idx[u]])
+
+                    cands.sort(key=lambda u: (-edges(u),
+                                              -len(by_uuid[u]), u))
+                else:
+                    cands.sort(key=lambda u: (-len(by_uuid[u]), u))
+                u = cands[0]
+                chosen_uuids.add(u)
+                while need > 0 and by_uuid[u]:
+                    chosen.append(by_uuid[u].pop(0))
+                    need -= 1
+                if not by_uuid[u]:
+                    del by_uuid[u]
+            resp.container_responses.add(deviceIDs=chosen)
+        return resp
 
     def Allocate(self, request, context):
         node = self.cfg.node_name

@@ -453,3 +453,40 @@ class TestReconciliation:
         plugin.reconcile(set())  # pod gone
         assert not d.exists()
         assert plugin.cumask.used_count(uuid) == 0
+
+
+class TestPreferredAllocation:
+    """xGMI-aligned GetPreferredAllocation (live, unlike the reference's
+    commented-out verb, server.go:270-285 / rm/allocate.go:44-121)."""
+
+    def _req(self, avail, size, must=()):
+        req = dp.PreferredAllocationRequest()
+        req.container_requests.add(
+            available_deviceIDs=avail, must_include_deviceIDs=list(must),
+            allocation_size=size)
+        return req
+
+    def test_fractional_packs_one_gpu(self, plugin_env):
+        plugin, kubelet, client, rm, cfg = plugin_env
+        u0, u1 = rm.gpus[0].uuid, rm.gpus[1].uuid
+        avail = [f"{u0}-0", f"{u1}-0", f"{u0}-1", f"{u1}-1"]
+        resp = plugin.GetPreferredAllocation(self._req(avail, 2), None)
+        got = list(resp.container_responses[0].deviceIDs)
+        assert len(got) == 2
+        assert len({ResourceManager.uuid_of_fake(f) for f in got}) == 1
+
+    def test_must_include_respected_and_packed(self, plugin_env):
+        plugin, kubelet, client, rm, cfg = plugin_env
+        u0, u1 = rm.gpus[0].uuid, rm.gpus[1].uuid
+        avail = [f"{u0}-1", f"{u1}-0", f"{u1}-1"]
+        resp = plugin.GetPreferredAllocation(
+            self._req(avail, 2, must=[f"{u1}-2"]), None)
+        got = list(resp.container_responses[0].deviceIDs)
+        assert got[0] == f"{u1}-2"
+        # packs onto u1 rather than spreading to u0
+        assert set(got) == {f"{u1}-2", f"{u1}-0"} or set(got) == {f"{u1}-2", f"{u1}-1"}
+
+    def test_options_advertise_preferred(self, plugin_env):
+        plugin, kubelet, client, rm, cfg = plugin_env
+        opts = plugin.GetDevicePluginOptions(dp.Empty(), None)
+        assert opts.get_preferred_allocation_available
