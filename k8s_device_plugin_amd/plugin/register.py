@@ -11,9 +11,25 @@ import threading
 import time
 
 from ..device.amd import HANDSHAKE_ANNO, REGISTER_ANNO
-from ..utils.codec import encode_node_devices
+from ..utils.codec import encode_node_devices, encode_node_xgmi
 from ..utils.kubeclient import KubeClient
+from ..utils.types import XGMI_ANNO
 from .rm import ResourceManager
+
+XGMI_LINK_TYPE = 11  # KFD io_link type for xGMI (topology.py)
+
+
+def xgmi_adjacency(gpus):
+    """{uuid: [xGMI peer uuids]} from the KFD io_links (7 p2p links/GPU on
+    an 8-GPU MI355X node)."""
+    node_to_uuid = {g.node_id: g.uuid for g in gpus}
+    adj = {}
+    for g in gpus:
+        peers = [node_to_uuid[n] for n, t in g.io_links.items()
+                 if t == XGMI_LINK_TYPE and n in node_to_uuid]
+        if peers:
+            adj[g.uuid] = peers
+    return adj
 
 log = logging.getLogger(__name__)
 
@@ -24,6 +40,9 @@ def register_once(client: KubeClient, node_name: str, rm: ResourceManager) -> No
         HANDSHAKE_ANNO: "Reported " + time.strftime("%Y-%m-%d %H:%M:%S"),
         REGISTER_ANNO: encode_node_devices(devices),
     }
+    adj = xgmi_adjacency(rm.gpus)
+    if adj:
+        annos[XGMI_ANNO] = encode_node_xgmi(adj)
     client.patch_node_annotations(node_name, annos)
 
 

@@ -22,9 +22,10 @@ from typing import Dict, List, Optional, Tuple
 
 from ..device import KNOWN_DEVICES, get_devices, init_devices
 from ..utils import nodelock
-from ..utils.codec import decode_node_devices, encode_pod_devices, decode_pod_devices
+from ..utils.codec import decode_node_xgmi, decode_node_devices, encode_pod_devices, decode_pod_devices
 from ..utils.kubeclient import KubeClient
 from ..utils.types import (
+    XGMI_ANNO,
     ASSIGNED_NODE_ANNO,
     ASSIGNED_TIME_ANNO,
     BIND_PHASE_ALLOCATING,
@@ -120,6 +121,12 @@ class Scheduler:
                 for index, d in enumerate(nodedevices):
                     d.index = index
                     info.devices.append(d)
+                xgmi_anno = node.annotations.get(XGMI_ANNO)
+                if xgmi_anno:
+                    try:
+                        info.xgmi = decode_node_xgmi(xgmi_anno)
+                    except Exception as e:
+                        log.warning("bad xgmi anno on %s: %s", node.name, e)
                 self.node_manager.add_node(node.name, info)
         self.get_nodes_usage(node_names)
 
@@ -167,7 +174,7 @@ class Scheduler:
             overall: Dict[str, NodeUsage] = {}
             failed: Dict[str, str] = {}
             for node_id, node in self.node_manager.list_nodes().items():
-                usage = NodeUsage()
+                usage = NodeUsage(xgmi=node.xgmi)
                 for d in node.devices:
                     usage.devices.append(
                         DeviceUsage(

@@ -20,12 +20,17 @@ the enforcement layer, see ops/cumask.py).
 """
 from __future__ import annotations
 
+import itertools
 import logging
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Tuple
 
 from ..device import get_devices
 from ..utils.types import (
+    BEST_EFFORT,
+    GUARANTEED,
+    RESTRICTED,
+    SCHEDULER_POLICY_ANNO,
     ContainerDevice,
     ContainerDeviceRequest,
     ContainerDeviceRequests,
@@ -41,6 +46,8 @@ log = logging.getLogger(__name__)
 @dataclass
 class NodeUsage:
     devices: List[DeviceUsage] = field(default_factory=list)
+    # uuid -> xGMI peer uuids (empty when the node has not advertised it)
+    xgmi: Dict[str, List[str]] = field(default_factory=dict)
 
 
 @dataclass
@@ -64,25 +71,74 @@ def check_type(
     return False, False
 
 
+def _xgmi_select(
+    eligible: List[Tuple[int, DeviceUsage, int]],
+    nums: int,
+    policy: str,
+    xgmi: Dict[str, List[str]],
+) -> Optional[List[Tuple[int, DeviceUsage, int]]]:
+    """Pick ``nums`` of the eligible devices maximizing xGMI edges.
+
+    MI355X-native multi-GPU placement (the MLU ring-allocator analog,
+    reference mlu/allocator/spider.go policies): ``guaranteed`` requires a
+    full xGMI clique, ``restricted`` one NUMA node, ``best-effort`` takes
+    the most-connected subset.  Ties keep the greedy (busiest-first) order.
+    """
+    peer = {u: set(p) for u, p in xgmi.items()}
+
+    def edges(combo):
+        uuids = [e[1].id for e in combo]
+        return sum(1 for i in range(len(uuids)) for j in range(i + 1, len(uuids))
+                   if uuids[j] in peer.get(uuids[i], ()))
+
+    best = None
+    max_edges = nums * (nums - 1) // 2
+    for order, combo in enumerate(itertools.combinations(eligible, nums)):
+        if policy == GUARANTEED and edges(combo) != max_edges:
+            continue
+        if policy == RESTRICTED and len({e[1].numa for e in combo}) > 1:
+            continue
+        key = (-edges(combo), order)
+        if best is None or key < best[0]:
+            best = (key, list(combo))
+    return best[1] if best else None
+
+
 def fit_in_certain_device(
     node: NodeUsage,
     request: ContainerDeviceRequest,
     annos: Dict[str, str],
 ) -> Tuple[bool, Dict[str, List[ContainerDevice]]]:
-    """score.go:86-157."""
+    """score.go:86-157 (+ xGMI-aware multi-GPU subset selection)."""
     nums = request.nums
+    policy = annos.get(SCHEDULER_POLICY_ANNO, BEST_EFFORT)
+    # topology mode: gather every eligible device, then pick the subset;
+    # otherwise stop at the first fit exactly like the reference
+    topo_mode = nums > 1 and bool(node.xgmi)
     prevnuma = -1
+    eligible: List[Tuple[int, DeviceUsage, int]] = []  # (idx, dev, memreq)
     tmp_devs: Dict[str, List[ContainerDevice]] = {}
+
+    def emit(selection):
+        tmp_devs[request.type] = [
+            ContainerDevice(idx=i, uuid=d.id, type=request.type,
+                            usedmem=memreq, usedcores=request.coresreq)
+            for i, d, memreq in selection
+        ]
+        return True, tmp_devs
+
     for i in range(len(node.devices) - 1, -1, -1):
         d = node.devices[i]
         found, numa_assert = check_type(annos, d, request)
         if not found:
             continue
         if numa_assert and prevnuma != d.numa:
-            # restart the pick inside the new NUMA domain
-            nums = request.nums
+            # domain boundary: if the previous NUMA domain already holds
+            # enough devices, select within it
+            if len(eligible) >= nums:
+                break
             prevnuma = d.numa
-            tmp_devs = {}
+            eligible = []
         if d.count <= d.used:
             continue
         if request.coresreq > 100:
@@ -103,20 +159,18 @@ def fit_in_certain_device(
         # a core=0 job can't land on an already core-full card
         if d.totalcore != 0 and d.usedcores == d.totalcore and request.coresreq == 0:
             continue
-        if nums > 0:
-            nums -= 1
-            tmp_devs.setdefault(request.type, []).append(
-                ContainerDevice(
-                    idx=i,
-                    uuid=d.id,
-                    type=request.type,
-                    usedmem=memreq,
-                    usedcores=request.coresreq,
-                )
-            )
-        if nums == 0:
-            return True, tmp_devs
-    return False, tmp_devs
+        eligible.append((i, d, memreq))
+        if not topo_mode and len(eligible) == nums:
+            return emit(eligible)
+
+    if len(eligible) < nums:
+        return False, tmp_devs
+    if topo_mode:
+        selection = _xgmi_select(eligible, nums, policy, node.xgmi)
+        if selection is None:
+            return False, tmp_devs  # policy unsatisfiable on this node
+        return emit(selection)
+    return emit(eligible[:nums])
 
 
 def fit_in_devices(

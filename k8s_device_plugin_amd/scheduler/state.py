@@ -17,6 +17,8 @@ from ..utils.types import DeviceInfo, PodDevices, PodInfo
 class SchedNodeInfo:
     id: str
     devices: List[DeviceInfo] = field(default_factory=list)
+    # uuid -> xGMI peer uuids (from amd.io/node-xgmi; may be empty)
+    xgmi: Dict[str, List[str]] = field(default_factory=dict)
 
 
 @dataclass

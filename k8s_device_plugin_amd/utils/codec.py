@@ -128,3 +128,27 @@ def _to_int(s: str) -> int:
         return int(s)
     except ValueError:
         return 0
+
+
+# ---------------------------------------------------------------------------
+# xGMI adjacency wire codec (MI355X-native addition: lets the scheduler make
+# topology-aware multi-GPU picks; no reference analog — the NVIDIA aligned
+# allocator is node-local and dormant, rm/allocate.go:44-64)
+# ---------------------------------------------------------------------------
+def encode_node_xgmi(adj: "Dict[str, List[str]]") -> str:
+    """{uuid: [peer uuids]} -> "uuidA:uuidB|uuidC;uuidD:..;" """
+    parts = []
+    for uuid in sorted(adj):
+        parts.append(f"{uuid}:{'|'.join(sorted(adj[uuid]))}")
+    return ";".join(parts) + (";" if parts else "")
+
+
+def decode_node_xgmi(s: str) -> "Dict[str, List[str]]":
+    out: Dict[str, List[str]] = {}
+    for part in s.split(";"):
+        part = part.strip()
+        if not part:
+            continue
+        uuid, _, peers = part.partition(":")
+        out[uuid] = [p for p in peers.split("|") if p]
+    return out

@@ -34,6 +34,8 @@ SUPPORT_DEVICES: Dict[str, str] = {}
 
 # Node lock annotation.  Reference: pkg/util/nodelock/nodelock.go:18.
 NODE_LOCK_ANNO = "amd.io/mutex.lock"
+XGMI_ANNO = "amd.io/node-xgmi"
+SCHEDULER_POLICY_ANNO = "amd.com/gpu-scheduler-policy"
 NODE_LOCK_EXPIRE_SECONDS = 300.0  # 5 min auto-expiry (nodelock.go:96-103)
 
 DEVICE_LIMIT = 100  # max devices per request (types.go:41)
