@@ -150,6 +150,82 @@ static int read_cu_occupancy_percent(int dev) {
   return pct > 100 ? 100 : pct;
 }
 
+/* Self-attributed GPU time from DRM fdinfo: amdgpu exports per-client
+ * engine time ("drm-engine-gfx"/"drm-engine-compute" in ns) for every drm
+ * fd.  delta(engine)/delta(wall) is THIS process's share of the GPU — a
+ * smooth, time-integrated signal (vs the bouncing instantaneous
+ * cu_occupancy), which is what makes 10 co-located pods converge fairly.
+ * Unique drm-client-ids are deduped (one client can be mapped by several
+ * fds). */
+#define MAX_DRM_CLIENTS 32
+static uint64_t read_self_engine_ns(void) {
+  DIR *dir = opendir("/proc/self/fdinfo");
+  if (!dir) return 0;
+  uint64_t client_ids[MAX_DRM_CLIENTS];
+  uint64_t client_ns[MAX_DRM_CLIENTS];
+  int nclients = 0;
+  struct dirent *de;
+  while ((de = readdir(dir)) != NULL) {
+    if (de->d_name[0] == '.') continue;
+    char path[300];
+    snprintf(path, sizeof(path), "/proc/self/fdinfo/%s", de->d_name);
+    FILE *f = fopen(path, "r");
+    if (!f) continue;
+    char line[256];
+    uint64_t cid = 0, ns = 0;
+    int is_drm = 0;
+    while (fgets(line, sizeof(line), f)) {
+      unsigned long long v;
+      if (sscanf(line, "drm-client-id:%llu", &v) == 1) {
+        cid = v;
+        is_drm = 1;
+      } else if (sscanf(line, "drm-engine-gfx:%llu", &v) == 1 ||
+                 sscanf(line, "drm-engine-compute:%llu", &v) == 1) {
+        ns += v;
+      }
+    }
+    fclose(f);
+    if (!is_drm) continue;
+    int found = -1;
+    for (int i = 0; i < nclients; i++)
+      if (client_ids[i] == cid) { found = i; break; }
+    if (found < 0 && nclients < MAX_DRM_CLIENTS) {
+      client_ids[nclients] = cid;
+      client_ns[nclients] = ns;
+      nclients++;
+    } else if (found >= 0 && ns > client_ns[found]) {
+      client_ns[found] = ns; /* same client via several fds: take max */
+    }
+  }
+  closedir(dir);
+  uint64_t total = 0;
+  for (int i = 0; i < nclients; i++) total += client_ns[i];
+  return total;
+}
+
+static uint64_t g_prev_engine_ns = 0;
+static uint64_t g_prev_engine_wall = 0;
+
+/* percent of the GPU this process used since the last call; -1 if fdinfo
+ * has no drm clients (no device open yet, or kernel without fdinfo stats) */
+static int read_self_util_percent(void) {
+  uint64_t ns = read_self_engine_ns();
+  uint64_t now = now_ns();
+  if (g_prev_engine_wall == 0 || ns < g_prev_engine_ns) {
+    g_prev_engine_ns = ns;
+    g_prev_engine_wall = now;
+    return -1;
+  }
+  uint64_t dwall = now - g_prev_engine_wall;
+  if (dwall < 100ull * 1000 * 1000) return -2; /* window too small: keep last */
+  uint64_t dns = ns - g_prev_engine_ns;
+  g_prev_engine_ns = ns;
+  g_prev_engine_wall = now;
+  if (ns == 0) return -1;
+  int pct = (int)(dns * 100 / dwall);
+  return pct > 100 ? 100 : pct;
+}
+
 static int read_busy_percent(int dev) {
   const char *util_file = getenv("VGPU_UTIL_FILE");
   char buf[32] = {0};
@@ -182,22 +258,39 @@ static void refill(vgpu_region_t *r, uint64_t now) {
   double dt = (double)(now - last) / NSEC;
   if (dt > 1.0) dt = 1.0;
   double fixed = fixed_rate();
+  /* fdinfo engine time is whole-process (all devices); read it once per
+   * tick, outside the device loop */
+  int self_util = getenv("VGPU_UTIL_FILE") ? -3 : read_self_util_percent();
   for (int d = 0; d < VGPU_MAX_DEVICES; d++) {
     uint64_t lim = r->sm_limit[d];
     if (lim == 0 || lim >= 100) continue;
     double base = fixed > 0 ? fixed : RATE_FULL * (double)lim / 100.0;
     if (fixed <= 0) {
       /* utilization feedback: converge this container's measured
-       * utilization on the limit.  Own-process cu_occupancy first (correct
-       * attribution under co-location); device busy% as fallback. */
-      int util = read_cu_occupancy_percent(d);
-      if (util < 0) util = read_busy_percent(d);
+       * utilization on the limit.  Signal priority: test fixture file >
+       * fdinfo self engine time (smooth + correctly attributed under
+       * co-location) > own-process cu_occupancy > device busy%. */
+      int util;
+      if (self_util == -3)
+        util = read_busy_percent(d);       /* VGPU_UTIL_FILE test fixture */
+      else if (self_util == -2)
+        util = -1;                         /* window too small: refill only */
+      else if (self_util < 0) {
+        util = read_cu_occupancy_percent(d);
+        if (util < 0) util = read_busy_percent(d);
+      } else {
+        util = self_util;
+      }
       if (g_rate_scale[d] == 0) g_rate_scale[d] = SCALE_INIT;
       if (util >= 0) {
-        if ((uint64_t)util > lim)
-          g_rate_scale[d] *= SCALE_DOWN;
-        else if ((uint64_t)util < lim * 9 / 10)
-          g_rate_scale[d] *= SCALE_UP;
+        /* proportional controller: gentle, symmetric, converges fairly
+         * across co-located pods (the old x0.5/x1.06 bang-bang punished
+         * whoever sampled a busy instant first — 25x spread at 10 pods) */
+        double err = ((double)lim - (double)util) / (double)lim;
+        double adj = 1.0 + 0.15 * err;
+        if (adj < 0.5) adj = 0.5;
+        if (adj > 1.5) adj = 1.5;
+        g_rate_scale[d] *= adj;
         if (g_rate_scale[d] < SCALE_MIN) g_rate_scale[d] = SCALE_MIN;
         if (g_rate_scale[d] > SCALE_MAX) g_rate_scale[d] = SCALE_MAX;
       }
