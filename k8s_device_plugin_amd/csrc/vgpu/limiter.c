@@ -53,11 +53,17 @@ static int g_cards_resolved = 0;
 /* Feedback shape: start conservative and clamp fast.  Kernel-heavy
  * workloads (few big launches) are bound by the busy%/occupancy loop;
  * launch-bound workloads by the credit rate itself. */
-#define SCALE_INIT 0.10
-#define SCALE_DOWN 0.50
-#define SCALE_UP 1.06
+#define SCALE_INIT 0.25
 #define SCALE_MIN 0.002
 #define SCALE_MAX 50.0
+/* EMA time constant for the utilization estimate: instantaneous
+ * cu_occupancy aliases with the pod's own activity (a pod that is running
+ * when it samples reads ~100, an idle one reads 0), so the controller
+ * must act on the smoothed duty cycle, not single samples. alpha=0.05 at
+ * 50 ms ticks ~= 1 s window. */
+#define UTIL_EMA_ALPHA 0.05
+static double g_util_ema[VGPU_MAX_DEVICES];
+static int g_util_ema_primed[VGPU_MAX_DEVICES];
 
 static uint64_t now_ns(void) {
   struct timespec ts;
@@ -273,23 +279,30 @@ static void refill(vgpu_region_t *r, uint64_t now) {
       int util;
       if (self_util == -3)
         util = read_busy_percent(d);       /* VGPU_UTIL_FILE test fixture */
-      else if (self_util == -2)
-        util = -1;                         /* window too small: refill only */
-      else if (self_util < 0) {
+      else if (self_util >= 0)
+        util = self_util;                  /* fdinfo engine time (if kernel
+                                            * exposes it for KFD queues) */
+      else {
         util = read_cu_occupancy_percent(d);
-        if (util < 0) util = read_busy_percent(d);
-      } else {
-        util = self_util;
+        if (util < 0 && self_util != -2) util = read_busy_percent(d);
       }
       if (g_rate_scale[d] == 0) g_rate_scale[d] = SCALE_INIT;
       if (util >= 0) {
-        /* proportional controller: gentle, symmetric, converges fairly
-         * across co-located pods (the old x0.5/x1.06 bang-bang punished
-         * whoever sampled a busy instant first — 25x spread at 10 pods) */
-        double err = ((double)lim - (double)util) / (double)lim;
-        double adj = 1.0 + 0.15 * err;
-        if (adj < 0.5) adj = 0.5;
-        if (adj > 1.5) adj = 1.5;
+        if (!g_util_ema_primed[d]) {
+          g_util_ema[d] = util;
+          g_util_ema_primed[d] = 1;
+        } else {
+          g_util_ema[d] = (1.0 - UTIL_EMA_ALPHA) * g_util_ema[d] +
+                          UTIL_EMA_ALPHA * (double)util;
+        }
+        /* proportional controller on the smoothed estimate: gentle and
+         * symmetric so co-located pods converge to equal shares (bang-bang
+         * or raw-sample control punished whichever pod happened to sample
+         * its own busy instant — 25x spread at 10 pods) */
+        double err = ((double)lim - g_util_ema[d]) / (double)lim;
+        double adj = 1.0 + 0.10 * err;
+        if (adj < 0.85) adj = 0.85;
+        if (adj > 1.15) adj = 1.15;
         g_rate_scale[d] *= adj;
         if (g_rate_scale[d] < SCALE_MIN) g_rate_scale[d] = SCALE_MIN;
         if (g_rate_scale[d] > SCALE_MAX) g_rate_scale[d] = SCALE_MAX;
