@@ -1,0 +1,69 @@
+/* Fake libamd_smi for amd-smi spoofing tests (fakehip/fakehsa pattern).
+ * Two fake GPUs, 288 GiB each, fixed "physical" usage of 200 GiB. */
+#define _GNU_SOURCE
+#include <stdint.h>
+#include <stdlib.h>
+
+typedef int amdsmi_status_t;
+typedef void *amdsmi_processor_handle;
+typedef void *amdsmi_socket_handle;
+typedef int amdsmi_memory_type_t;
+typedef struct {
+  uint32_t vram_total;
+  uint32_t vram_used;
+  uint32_t reserved[2];
+} amdsmi_vram_usage_t;
+
+static char g_devs[2]; /* handles = their addresses */
+
+amdsmi_status_t amdsmi_init(uint64_t flags) { (void)flags; return 0; }
+amdsmi_status_t amdsmi_shut_down(void) { return 0; }
+
+amdsmi_status_t amdsmi_get_socket_handles(uint32_t *count,
+                                          amdsmi_socket_handle *handles) {
+  if (count) {
+    if (handles && *count >= 1) handles[0] = (void *)0x5;
+    *count = 1;
+  }
+  return 0;
+}
+
+amdsmi_status_t amdsmi_get_processor_handles(amdsmi_socket_handle socket,
+                                             uint32_t *count,
+                                             amdsmi_processor_handle *handles) {
+  (void)socket;
+  if (!count) return 1;
+  if (handles && *count >= 2) {
+    handles[0] = &g_devs[0];
+    handles[1] = &g_devs[1];
+  }
+  *count = 2;
+  return 0;
+}
+
+amdsmi_status_t amdsmi_get_gpu_memory_total(amdsmi_processor_handle h,
+                                            amdsmi_memory_type_t type,
+                                            uint64_t *total) {
+  (void)h;
+  if (type != 0 || !total) return 1;
+  *total = 288ULL << 30;
+  return 0;
+}
+
+amdsmi_status_t amdsmi_get_gpu_memory_usage(amdsmi_processor_handle h,
+                                            amdsmi_memory_type_t type,
+                                            uint64_t *used) {
+  (void)h;
+  if (type != 0 || !used) return 1;
+  *used = 200ULL << 30;
+  return 0;
+}
+
+amdsmi_status_t amdsmi_get_gpu_vram_usage(amdsmi_processor_handle h,
+                                          amdsmi_vram_usage_t *info) {
+  (void)h;
+  if (!info) return 1;
+  info->vram_total = 288u << 10; /* MB */
+  info->vram_used = 200u << 10;
+  return 0;
+}

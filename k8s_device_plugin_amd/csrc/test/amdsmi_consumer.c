@@ -1,0 +1,47 @@
+/* PLT-linked amd-smi-style consumer: enumerates processors and prints the
+ * memory view per device — run under LD_PRELOAD to verify quota spoofing. */
+#include <stdint.h>
+#include <stdio.h>
+
+typedef int amdsmi_status_t;
+typedef void *amdsmi_processor_handle;
+typedef void *amdsmi_socket_handle;
+typedef struct {
+  uint32_t vram_total;
+  uint32_t vram_used;
+  uint32_t reserved[2];
+} amdsmi_vram_usage_t;
+
+extern amdsmi_status_t amdsmi_init(uint64_t);
+extern amdsmi_status_t amdsmi_get_processor_handles(amdsmi_socket_handle,
+                                                    uint32_t *,
+                                                    amdsmi_processor_handle *);
+extern amdsmi_status_t amdsmi_get_gpu_memory_total(amdsmi_processor_handle,
+                                                   int, uint64_t *);
+extern amdsmi_status_t amdsmi_get_gpu_memory_usage(amdsmi_processor_handle,
+                                                   int, uint64_t *);
+extern amdsmi_status_t amdsmi_get_gpu_vram_usage(amdsmi_processor_handle,
+                                                 amdsmi_vram_usage_t *);
+
+int main(void) {
+  amdsmi_init(0);
+  amdsmi_processor_handle hs[16];
+  uint32_t n = 16;
+  if (amdsmi_get_processor_handles(NULL, &n, hs) != 0) {
+    printf("enumerate failed\n");
+    return 1;
+  }
+  for (uint32_t i = 0; i < n; i++) {
+    uint64_t total = 0, used = 0;
+    amdsmi_vram_usage_t vu = {0};
+    amdsmi_get_gpu_memory_total(hs[i], 0, &total);
+    amdsmi_get_gpu_memory_usage(hs[i], 0, &used);
+    amdsmi_get_gpu_vram_usage(hs[i], &vu);
+    printf("{\"dev\":%u,\"total\":%llu,\"used\":%llu,"
+           "\"vram_total_mb\":%u,\"vram_used_mb\":%u}\n",
+           i, (unsigned long long)total, (unsigned long long)used,
+           vu.vram_total, vu.vram_used);
+  }
+  fflush(stdout);
+  return 0;
+}

@@ -80,6 +80,18 @@ void *vgpu_real_rsmi_handle(void) {
 
 __thread int vgpu_tls_passthrough = 0;
 
+static void *g_real_amdsmi = NULL;
+
+void *vgpu_real_amdsmi(const char *sym) {
+  if (!g_real_amdsmi) {
+    static const char *const names[] = {"libamd_smi.so.26", "libamd_smi.so.25",
+                                        "libamd_smi.so", NULL};
+    g_real_amdsmi = open_real(ENV_REAL_AMDSMI, names, "libamd_smi");
+  }
+  if (!g_real_amdsmi) return NULL;
+  return dlsym(g_real_amdsmi, sym);
+}
+
 static void *g_real_hsa = NULL;
 
 void *vgpu_real_hsa(const char *sym) {
@@ -128,7 +140,7 @@ static void *g_self_handle = NULL; /* handle apps got from a redirect */
 void *dlopen(const char *filename, int flags) {
   if (filename && !tls_no_redirect && !vgpu_control_disabled()) {
     if (strstr(filename, "libamdhip64") || strstr(filename, "librocm_smi64") ||
-        strstr(filename, "libhsa-runtime64")) {
+        strstr(filename, "libhsa-runtime64") || strstr(filename, "libamd_smi")) {
       const char *self = self_path();
       if (self) {
         vgpu_log(VGPU_INFO, "redirecting dlopen(%s) to %s", filename, self);
@@ -140,6 +152,8 @@ void *dlopen(const char *filename, int flags) {
             vgpu_real_hip("hipGetDeviceCount");
           else if (strstr(filename, "libhsa-runtime64"))
             vgpu_real_hsa("hsa_init");
+          else if (strstr(filename, "libamd_smi"))
+            vgpu_real_amdsmi("amdsmi_init");
           else
             vgpu_real_rsmi_handle();
           return h;
@@ -174,9 +188,11 @@ void *dlsym(void *handle, const char *symbol) {
   void *p = real(handle, symbol);
   if (!p && symbol && g_self_handle && handle == g_self_handle) {
     tls_no_redirect++;
-    if (strncmp(symbol, "rsmi_", 5) == 0 || strncmp(symbol, "amdsmi_", 7) == 0) {
+    if (strncmp(symbol, "rsmi_", 5) == 0) {
       void *h = vgpu_real_rsmi_handle();
       if (h) p = real(h, symbol);
+    } else if (strncmp(symbol, "amdsmi_", 7) == 0) {
+      p = vgpu_real_amdsmi(symbol);
     } else if (strncmp(symbol, "hsa_", 4) == 0) {
       p = vgpu_real_hsa(symbol);
     } else {

@@ -25,6 +25,10 @@ typedef int rsmi_status_t;
 #define RSMI_STATUS_SUCCESS 0
 typedef int rsmi_memory_type_t; /* RSMI_MEM_TYPE_VRAM = 0 */
 
+static int rsmi_index_to_vdev(uint32_t dv_ind);
+
+int vgpu_smi_index_to_vdev(uint32_t idx) { return rsmi_index_to_vdev(idx); }
+
 static int rsmi_index_to_vdev(uint32_t dv_ind) {
   static int parsed = 0;
   static int map[VGPU_MAX_DEVICES];

@@ -45,6 +45,7 @@ extern "C" {
 #define ENV_REAL_HIP "VGPU_REAL_HIP_PATH"             /* test hook: fake lib */
 #define ENV_REAL_RSMI "VGPU_REAL_RSMI_PATH"
 #define ENV_REAL_HSA "VGPU_REAL_HSA_PATH"
+#define ENV_REAL_AMDSMI "VGPU_REAL_AMDSMI_PATH"
 #define ENV_DEVICE_UUIDS "VGPU_DEVICE_UUIDS"          /* comma list, monitor correlation */
 
 /* memory accounting split, per device per process (reference ABI:
@@ -113,6 +114,8 @@ void vgpu_limiter_gate(int dev, uint64_t workgroups); /* blocks when throttled *
 void *vgpu_real_hip(const char *sym);   /* resolve real libamdhip64 symbol */
 void *vgpu_real_rsmi_handle(void);
 void *vgpu_real_hsa(const char *sym);   /* resolve real libhsa-runtime64 symbol */
+void *vgpu_real_amdsmi(const char *sym); /* resolve real libamd_smi symbol */
+int vgpu_smi_index_to_vdev(uint32_t idx); /* smi.c: tool device idx -> vdev */
 /* >0 while inside one of our own wrappers: lower-layer hooks (hsa.c) must
  * pass through, not double-count (the HIP runtime allocates via HSA). */
 extern __thread int vgpu_tls_passthrough;

@@ -1,0 +1,64 @@
+"""amd-smi (libamd_smi) quota spoofing against the fake library.
+
+The reference spoofs nvidia-smi via ~260 nvml hooks;
+rocm-smi is covered by smi.c and amd-smi — the modern CLI — by amdsmi.c.
+"""
+import json
+import os
+import subprocess
+from pathlib import Path
+
+import pytest
+
+CSRC = Path(__file__).resolve().parent.parent / "k8s_device_plugin_amd" / "csrc"
+LIBVGPU = CSRC / "libvgpu-hip.so"
+FAKEDIR = CSRC / "fakeamdsmi"
+CONSUMER = CSRC / "test" / "amdsmi_consumer"
+
+GIB = 1 << 30
+MIB = 1 << 20
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    if not (LIBVGPU.exists() and CONSUMER.exists()):
+        subprocess.run(["make"], cwd=CSRC, check=True, capture_output=True)
+
+
+def run_consumer(cache, limit=None, preload=True, extra=None):
+    env = dict(os.environ)
+    env["LD_LIBRARY_PATH"] = str(FAKEDIR)
+    if preload:
+        env["LD_PRELOAD"] = str(LIBVGPU)
+    env["VGPU_DEVICE_MEMORY_SHARED_CACHE"] = str(cache)
+    env["VGPU_REAL_AMDSMI_PATH"] = str(FAKEDIR / "libamd_smi.so")
+    env["VGPU_REAL_HIP_PATH"] = str(CSRC / "fakehip" / "libamdhip64.so")
+    if limit:
+        env["VGPU_DEVICE_MEMORY_LIMIT"] = limit
+    env.update(extra or {})
+    out = subprocess.run([str(CONSUMER)], env=env, capture_output=True,
+                         text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    return [json.loads(l) for l in out.stdout.splitlines() if l.startswith("{")]
+
+
+def test_amdsmi_reports_quota(tmp_path):
+    devs = run_consumer(tmp_path / "r.cache", limit="73728m")
+    assert len(devs) == 2
+    # device 0 is the container's vGPU: totals clamp to the 72 GiB quota
+    assert devs[0]["total"] == 72 * GIB
+    assert devs[0]["used"] == 0          # nothing allocated by this container
+    assert devs[0]["vram_total_mb"] == 73728
+    assert devs[0]["vram_used_mb"] == 0
+
+
+def test_amdsmi_passthrough_without_preload(tmp_path):
+    devs = run_consumer(tmp_path / "r.cache", limit="73728m", preload=False)
+    assert devs[0]["total"] == 288 * GIB
+    assert devs[0]["used"] == 200 * GIB
+
+
+def test_amdsmi_disable_control(tmp_path):
+    devs = run_consumer(tmp_path / "r.cache", limit="73728m",
+                        extra={"VGPU_DISABLE_CONTROL": "1"})
+    assert devs[0]["total"] == 288 * GIB
