@@ -290,3 +290,37 @@ class TestSchedulerEndToEnd:
         assert self.sched.node_manager.get_node("node1") is None
         assert self.client.get_node("node1").annotations[HANDSHAKE_ANNO].startswith(
             "Deleted_")
+
+
+class TestExtenderTLS:
+    """The webhook endpoint must serve HTTPS (reference scheduler runs the
+    extender behind --cert_file/--key_file, cmd/scheduler/main.go:52-56)."""
+
+    def test_filter_over_https(self, tmp_path):
+        import ssl
+        import subprocess as sp
+        import urllib.request
+
+        from k8s_device_plugin_amd.scheduler.core import Scheduler
+        from k8s_device_plugin_amd.scheduler.routes import ExtenderServer
+        from k8s_device_plugin_amd.utils.kubeclient import FakeKubeClient
+
+        cert = tmp_path / "tls.crt"
+        key = tmp_path / "tls.key"
+        sp.run(["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+                "-keyout", str(key), "-out", str(cert), "-days", "1",
+                "-subj", "/CN=127.0.0.1"], check=True, capture_output=True)
+        sched = Scheduler(FakeKubeClient())
+        server = ExtenderServer(sched, host="127.0.0.1", port=0,
+                                cert_file=str(cert), key_file=str(key))
+        server.start()
+        try:
+            ctx = ssl.create_default_context()
+            ctx.check_hostname = False
+            ctx.verify_mode = ssl.CERT_NONE
+            with urllib.request.urlopen(
+                    f"https://127.0.0.1:{server.port}/healthz",
+                    context=ctx, timeout=10) as resp:
+                assert resp.status == 200
+        finally:
+            server.stop()
