@@ -65,6 +65,8 @@ def main():
                    help="HBM percent per pod (default 100/pods)")
     p.add_argument("--cu-pct", type=int, default=0,
                    help="CU percent per pod (default 100/pods)")
+    p.add_argument("--no-cu-mask", action="store_true",
+                   help="skip the hard HSA_CU_MASK partition (soft limiter only)")
     p.add_argument("--worker", action="store_true", help=argparse.SUPPRESS)
     args = p.parse_args()
     if args.worker:
@@ -78,6 +80,18 @@ def main():
     quota_pct = args.quota_pct or max(1, 100 // args.pods)
     cu_pct = args.cu_pct or max(1, 100 // args.pods)
     quota_mib = total_mem * quota_pct // 100 // (1 << 20)
+
+    # hard partition: disjoint CU masks, exactly as the plugin injects at
+    # Allocate (ops/cumask.py) — 10 pods on 256 CUs -> 25 CUs each.  The
+    # soft token-bucket limiter stays on top (both layers, the production
+    # stack).  Fairness comes from the hardware partition.
+    total_cus = 256
+    chunk = total_cus // args.pods
+    masks = []
+    for i in range(args.pods):
+        lo = i * chunk
+        hi = lo + chunk - 1
+        masks.append(f"0:{lo}-{hi}")
 
     procs = []
     caches = []
@@ -94,6 +108,8 @@ def main():
             "VGPU_DEVICE_MEMORY_SHARED_CACHE": cache.name,
             "VGPU_DEVICE_UUIDS": f"GPU-density-{i}",
         })
+        if not args.no_cu_mask:
+            env["HSA_CU_MASK"] = masks[i]
         procs.append(subprocess.Popen(
             [sys.executable, __file__, "--worker", "--case", args.case,
              "--seconds", str(args.seconds)],
@@ -127,6 +143,7 @@ def main():
         "fairness_max_over_min": round(max(rates) / max(min(rates), 1e-9), 3),
         "quota_pct": quota_pct,
         "cu_pct": cu_pct,
+        "hard_cu_mask": not args.no_cu_mask,
         "case": args.case,
         "seconds": args.seconds,
     }))
