@@ -1,0 +1,154 @@
+"""Full-stack CPU integration: the complete pod lifecycle across all three
+protocols (SURVEY.md §1 data flows) with zero hardware —
+
+  plugin registers node annos -> scheduler ingests -> webhook mutates pod ->
+  /filter picks + pre-patches -> /bind locks + allocating -> kubelet
+  Allocate (gRPC over unix socket) injects env/mounts -> a PLT-linked HIP
+  consumer under LD_PRELOAD enforces the injected quota in the shared
+  region -> the monitor scans the region and exports the pod's usage.
+
+The reference has no test like this at any level (no e2e).
+"""
+import json
+import os
+import subprocess
+import time
+from concurrent import futures
+from pathlib import Path
+
+import grpc
+import pytest
+
+from k8s_device_plugin_amd.monitor.metrics import MonitorCollector, metrics_text
+from k8s_device_plugin_amd.monitor.pathmon import PathMonitor
+from k8s_device_plugin_amd.plugin.config import PluginConfig
+from k8s_device_plugin_amd.plugin.kfd import enumerate_gpus
+from k8s_device_plugin_amd.plugin.register import register_once
+from k8s_device_plugin_amd.plugin.rm import ResourceManager
+from k8s_device_plugin_amd.plugin.server import VGPUDevicePlugin
+from k8s_device_plugin_amd.proto import deviceplugin as dp
+from k8s_device_plugin_amd.scheduler.core import Scheduler
+from k8s_device_plugin_amd.scheduler.webhook import handle_admission_review
+from k8s_device_plugin_amd.utils.kubeclient import FakeKubeClient
+from k8s_device_plugin_amd.utils.types import (
+    BIND_PHASE_ANNO,
+    BIND_PHASE_SUCCESS,
+    ASSIGNED_NODE_ANNO,
+    NodeInfo,
+    PodInfo,
+)
+from tests.test_plugin import make_kfd_tree
+
+CSRC = Path(__file__).resolve().parent.parent / "k8s_device_plugin_amd" / "csrc"
+MIB = 1 << 20
+
+
+def test_pod_lifecycle_end_to_end(tmp_path):
+    client = FakeKubeClient()
+    client.add_node(NodeInfo(name="n1"))
+
+    # ---- node agent comes up and registers --------------------------------
+    topo, pci = make_kfd_tree(tmp_path, n_gpus=2)
+    rm = ResourceManager(enumerate_gpus(str(topo), str(pci)), split_count=4)
+    cfg = PluginConfig(
+        node_name="n1",
+        hook_path=str(tmp_path / "hook"),
+        plugin_socket_dir=str(tmp_path),
+        kubelet_socket=str(tmp_path / "kubelet.sock"),
+    )
+    plugin = VGPUDevicePlugin(cfg, rm, client)
+    plugin.serve()
+    register_once(client, "n1", rm)
+
+    # ---- scheduler ingests the node ---------------------------------------
+    sched = Scheduler(client)
+    sched.register_from_node_annotations_once()
+
+    # ---- user creates a pod; webhook mutates it ---------------------------
+    pod_obj = {
+        "kind": "Pod",
+        "metadata": {"name": "train-1", "namespace": "default",
+                     "uid": "uid-train-1"},
+        "spec": {"containers": [{
+            "name": "main",
+            "resources": {"limits": {
+                "amd.com/gpu": "1",
+                "amd.com/gpumem": "73728",
+                "amd.com/gpucores": "25",
+            }},
+        }]},
+    }
+    review = handle_admission_review(
+        {"apiVersion": "admission.k8s.io/v1", "kind": "AdmissionReview",
+         "request": {"uid": "rev-1", "object": pod_obj,
+                     "kind": {"kind": "Pod"}}})
+    assert review["response"]["allowed"]
+    import base64
+
+    patches = json.loads(base64.b64decode(review["response"]["patch"]))
+    assert any(p["path"] == "/spec/schedulerName" for p in patches)
+
+    pod = PodInfo.from_k8s(pod_obj)
+    client.add_pod(pod)
+
+    # ---- /filter then /bind ------------------------------------------------
+    fr = sched.filter(pod, ["n1"])
+    assert fr.node_names == ["n1"], fr
+    assert pod.annotations[ASSIGNED_NODE_ANNO] == "n1"
+    br = sched.bind("train-1", "default", "n1")
+    assert not br.error
+
+    # ---- kubelet calls Allocate -------------------------------------------
+    with grpc.insecure_channel(f"unix://{plugin.socket_path}") as ch:
+        stub = dp.DevicePluginClient(ch)
+        req = dp.AllocateRequest()
+        # kubelet hands over one fake-device id; the assignment in the
+        # annotations is authoritative for WHICH physical GPU
+        req.container_requests.add(devicesIDs=[f"{rm.gpus[0].uuid}-0"])
+        resp = stub.Allocate(req)
+    envs = dict(resp.container_responses[0].envs)
+    assert envs["VGPU_DEVICE_MEMORY_LIMIT_0"] == "73728m"
+    assert envs["VGPU_DEVICE_CU_LIMIT"] == "25"
+    assert "HSA_CU_MASK" in envs
+    assert client.get_pod("train-1").annotations[BIND_PHASE_ANNO] == BIND_PHASE_SUCCESS
+
+    # ---- the container runs: enforcement via the injected env -------------
+    ctr_dir = Path(cfg.hook_path) / "vgpu" / "containers" / "uid-train-1_main"
+    assert ctr_dir.is_dir()  # created at Allocate
+    cache = ctr_dir / "region.cache"
+    run_env = dict(os.environ)
+    run_env.update({
+        "LD_LIBRARY_PATH": str(CSRC / "fakehip"),
+        "LD_PRELOAD": str(CSRC / "libvgpu-hip.so"),
+        "VGPU_REAL_HIP_PATH": str(CSRC / "fakehip" / "libamdhip64.so"),
+        "VGPU_DEVICE_MEMORY_LIMIT_0": envs["VGPU_DEVICE_MEMORY_LIMIT_0"],
+        "VGPU_DEVICE_MEMORY_LIMIT": envs["VGPU_DEVICE_MEMORY_LIMIT_0"],
+        "VGPU_DEVICE_CU_LIMIT": envs["VGPU_DEVICE_CU_LIMIT"],
+        "VGPU_DEVICE_UUIDS": envs["VGPU_DEVICE_UUIDS"],
+        # the real mount maps the host cache dir; point the region there
+        "VGPU_DEVICE_MEMORY_SHARED_CACHE": str(cache),
+    })
+    out = subprocess.run(
+        [str(CSRC / "test" / "hip_consumer"),
+         "meminfo", "alloc", str(1024 * MIB),
+         "alloc", str(80000 * MIB), "meminfo", "sleep", "1000"],
+        env=run_env, capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    lines = [json.loads(l) for l in out.stdout.splitlines()]
+    assert lines[0]["total"] == 73728 * MIB     # quota view
+    assert lines[1]["err"] == 0                 # within quota
+    assert lines[2]["err"] == 2                 # beyond quota -> OOM
+    assert lines[3]["free"] == (73728 - 1024) * MIB
+
+    # ---- monitor sees the container ----------------------------------------
+    pathmon = PathMonitor(str(Path(cfg.hook_path) / "vgpu"))
+    pathmon.scan({"uid-train-1"})
+    # the consumer exited, but the region file persists with its ledger
+    entries = pathmon.live_regions()
+    assert len(entries) == 1
+    snap = entries[0].region.snapshot()
+    assert snap.limit[0] == 73728 * MIB
+    mtext = metrics_text(MonitorCollector(pathmon, rm.gpus)).decode()
+    assert "vGPU_device_memory_limit_in_bytes" in mtext
+
+    plugin.stop()
