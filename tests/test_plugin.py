@@ -490,3 +490,51 @@ class TestPreferredAllocation:
         plugin, kubelet, client, rm, cfg = plugin_env
         opts = plugin.GetDevicePluginOptions(dp.Empty(), None)
         assert opts.get_preferred_allocation_available
+
+
+class TestMultiContainerAllocate:
+    """Two GPU containers in one pod: each Allocate call consumes ONE
+    container's assignment (reference util.go:216-271 erase semantics)."""
+
+    def test_two_containers_two_calls(self, plugin_env):
+        plugin, kubelet, client, rm, cfg = plugin_env
+        u0, u1 = rm.gpus[0].uuid, rm.gpus[1].uuid
+        devs = [
+            [ContainerDevice(uuid=u0, type="AMD", usedmem=1024, usedcores=10)],
+            [ContainerDevice(uuid=u1, type="AMD", usedmem=2048, usedcores=20)],
+        ]
+        pod = PodInfo(
+            name="p2", uid="uid-p2",
+            containers=[
+                ContainerSpec(name="c0", limits={"amd.com/gpu": 1}),
+                ContainerSpec(name="c1", limits={"amd.com/gpu": 1}),
+            ],
+            annotations={
+                BIND_TIME_ANNO: "123",
+                BIND_PHASE_ANNO: BIND_PHASE_ALLOCATING,
+                ASSIGNED_NODE_ANNO: "node1",
+                IN_REQUEST_DEVICES["AMD"]: encode_pod_single_device(devs),
+            },
+        )
+        client.add_pod(pod)
+        client.patch_node_annotations(
+            "node1", {NODE_LOCK_ANNO: "2026-01-01T00:00:00Z"})
+        with grpc.insecure_channel(f"unix://{plugin.socket_path}") as ch:
+            stub = dp.DevicePluginClient(ch)
+            # kubelet allocates container by container (one request each)
+            req = dp.AllocateRequest()
+            req.container_requests.add(devicesIDs=[f"{u0}-0"])
+            r1 = stub.Allocate(req)
+            e1 = dict(r1.container_responses[0].envs)
+            assert e1["VGPU_DEVICE_MEMORY_LIMIT_0"] == "1024m"
+            # pod not yet fully consumed -> still allocating
+            assert client.get_pod("p2").annotations[BIND_PHASE_ANNO] == BIND_PHASE_ALLOCATING
+            req = dp.AllocateRequest()
+            req.container_requests.add(devicesIDs=[f"{u1}-0"])
+            r2 = stub.Allocate(req)
+            e2 = dict(r2.container_responses[0].envs)
+            assert e2["VGPU_DEVICE_MEMORY_LIMIT_0"] == "2048m"
+            assert e2["ROCR_VISIBLE_DEVICES"] == u1
+        # both consumed -> success + lock released
+        assert client.get_pod("p2").annotations[BIND_PHASE_ANNO] == BIND_PHASE_SUCCESS
+        assert NODE_LOCK_ANNO not in client.get_node("node1").annotations
