@@ -67,6 +67,10 @@ def main():
                    help="CU percent per pod (default 100/pods)")
     p.add_argument("--no-cu-mask", action="store_true",
                    help="skip the hard HSA_CU_MASK partition (soft limiter only)")
+    p.add_argument("--prewarm", action="store_true",
+                   help="run one unmasked pass first to populate the MIOpen "
+                        "find-db (keyed by conv config, not CU mask) so the "
+                        "masked pods skip autotuning")
     p.add_argument("--worker", action="store_true", help=argparse.SUPPRESS)
     args = p.parse_args()
     if args.worker:
@@ -80,6 +84,22 @@ def main():
     quota_pct = args.quota_pct or max(1, 100 // args.pods)
     cu_pct = args.cu_pct or max(1, 100 // args.pods)
     quota_mib = total_mem * quota_pct // 100 // (1 << 20)
+
+    if args.prewarm:
+        import torch as _t
+
+        sys.path.insert(0, str(REPO))
+        from k8s_device_plugin_amd.models import zoo as _zoo
+
+        _case = _zoo.CASES[args.case]
+        _model = _zoo.build(_case, _t.device("cuda", 0))
+        _batch = _zoo.synthetic_batch(_case, _t.device("cuda", 0))
+        for _ in range(3):
+            _zoo.step(_case, _model, _batch, None)
+        _t.cuda.synchronize()
+        del _model, _batch
+        _t.cuda.empty_cache()
+        print("prewarm done", flush=True)
 
     # hard partition: disjoint CU masks, exactly as the plugin injects at
     # Allocate (ops/cumask.py) — 10 pods on 256 CUs -> 25 CUs each.  The
