@@ -142,6 +142,7 @@ class Scheduler:
     def register_loop(self) -> None:
         while not self._stop.is_set():
             self.register_from_node_annotations_once()
+            self.rebuild_pod_cache()
             self._stop.wait(REGISTER_POLL_INTERVAL_S)
 
     def stop(self) -> None:
@@ -159,10 +160,24 @@ class Scheduler:
         self.pod_manager.add_pod(pod, node_id, devices)
 
     def rebuild_pod_cache(self) -> None:
-        for pod in self.client.list_pods():
+        """Reconcile the pod cache against the API: (re)ingest assigned
+        pods AND evict vanished/completed ones (the reference's informer
+        onAddPod/onDelPod pair, scheduler.go:73-126 — without eviction a
+        deleted pod's usage would leak until restart)."""
+        try:
+            pods = self.client.list_pods()
+        except Exception as e:
+            log.error("pod list failed: %s", e)
+            return
+        live = set()
+        for pod in pods:
             if pod.phase in ("Succeeded", "Failed"):
                 continue
+            live.add(pod.uid)
             self.ingest_assigned_pod(pod)
+        for cached in self.pod_manager.list_pods():
+            if cached.uid not in live:
+                self.pod_manager.del_pod_by_uid(cached.uid)
 
     # ------------------------------------------------------------------
     # Usage snapshot

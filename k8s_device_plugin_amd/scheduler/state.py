@@ -41,8 +41,12 @@ class NodeManager:
         with self._lock:
             cur = self.nodes.get(node_id)
             if cur is None:
-                self.nodes[node_id] = SchedNodeInfo(id=node_id, devices=list(info.devices))
+                self.nodes[node_id] = SchedNodeInfo(
+                    id=node_id, devices=list(info.devices),
+                    xgmi=dict(info.xgmi))
                 return
+            if info.xgmi:
+                cur.xgmi = dict(info.xgmi)
             for d in info.devices:
                 for existing in cur.devices:
                     if existing.id == d.id:
@@ -89,6 +93,10 @@ class PodManager:
     def del_pod(self, pod: PodInfo) -> None:
         with self._lock:
             self.pods.pop(pod.uid, None)
+
+    def del_pod_by_uid(self, uid: str) -> None:
+        with self._lock:
+            self.pods.pop(uid, None)
 
     def list_pods(self) -> List[SchedPodInfo]:
         with self._lock:

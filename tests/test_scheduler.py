@@ -324,3 +324,51 @@ class TestExtenderTLS:
                 assert resp.status == 200
         finally:
             server.stop()
+
+
+class TestPodCacheEviction:
+    """rebuild_pod_cache must evict deleted/completed pods (the reference's
+    informer onDelPod, scheduler.go:91-110) or their usage leaks forever."""
+
+    def test_deleted_pod_usage_released(self):
+        from k8s_device_plugin_amd.device.amd import REGISTER_ANNO, HANDSHAKE_ANNO
+        from k8s_device_plugin_amd.scheduler.core import Scheduler
+        from k8s_device_plugin_amd.utils.codec import (
+            encode_node_devices,
+            encode_pod_single_device,
+        )
+        from k8s_device_plugin_amd.utils.kubeclient import FakeKubeClient
+        from k8s_device_plugin_amd.utils.types import (
+            ASSIGNED_NODE_ANNO,
+            SUPPORT_DEVICES,
+            ContainerDevice,
+            DeviceInfo,
+            NodeInfo,
+            PodInfo,
+        )
+
+        client = FakeKubeClient()
+        client.add_node(NodeInfo(name="n1", annotations={
+            HANDSHAKE_ANNO: "Reported 2026-01-01 00:00:00",
+            REGISTER_ANNO: encode_node_devices([DeviceInfo(
+                id="GPU-x", count=10, devmem=294912, devcore=100,
+                type="AMD-Instinct-MI355X", numa=0, health=True, index=0)]),
+        }))
+        sched = Scheduler(client)
+        sched.register_from_node_annotations_once()
+        pod = PodInfo(name="p", uid="uid-p", node_name="n1", annotations={
+            ASSIGNED_NODE_ANNO: "n1",
+            SUPPORT_DEVICES["AMD"]: encode_pod_single_device(
+                [[ContainerDevice(uuid="GPU-x", type="AMD", usedmem=1000,
+                                  usedcores=50)]]),
+        })
+        client.add_pod(pod)
+        sched.rebuild_pod_cache()
+        usage, _ = sched.get_nodes_usage(["n1"])
+        assert usage["n1"].devices[0].usedmem == 1000
+        # pod deleted from the API -> next rebuild releases its usage
+        client.delete_pod("p")
+        sched.rebuild_pod_cache()
+        usage, _ = sched.get_nodes_usage(["n1"])
+        assert usage["n1"].devices[0].usedmem == 0
+        assert usage["n1"].devices[0].used == 0
