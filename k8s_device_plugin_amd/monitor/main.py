@@ -100,9 +100,12 @@ def main(argv=None):
                 live_uids = {e.pod_uid for e in pathmon.entries.values()}
         else:
             live_uids = {e.pod_uid for e in pathmon.entries.values()}
-        pathmon.scan(live_uids)
-        update_host_pids(pathmon)
-        feedback.observe_once()
+        try:
+            pathmon.scan(live_uids)
+            update_host_pids(pathmon)
+            feedback.observe_once()
+        except Exception:
+            log.exception("monitor tick failed")  # keep the loop alive
         time.sleep(args.interval)
 
 
