@@ -271,7 +271,16 @@ static void refill(vgpu_region_t *r, uint64_t now) {
     uint64_t lim = r->sm_limit[d];
     if (lim == 0 || lim >= 100) continue;
     double base = fixed > 0 ? fixed : RATE_FULL * (double)lim / 100.0;
-    if (fixed <= 0) {
+    /* monitor-arbitrated mode: a fresh node-level scale overrides the
+     * local feedback loop — every co-located container gets the SAME
+     * multiplier on its entitled share (proportional fairness) */
+    uint64_t mts = __atomic_load_n(&r->monitor_scale_ts_ns, __ATOMIC_RELAXED);
+    int monitor_fresh = fixed <= 0 && mts != 0 && now > mts &&
+                        now - mts < 2ULL * NSEC;
+    if (monitor_fresh) {
+      int64_t fp = __atomic_load_n(&r->monitor_scale_fp[d], __ATOMIC_RELAXED);
+      if (fp > 0) base *= (double)fp / 1e6;
+    } else if (fixed <= 0) {
       /* utilization feedback: converge this container's measured
        * utilization on the limit.  Signal priority: test fixture file >
        * fdinfo self engine time (smooth + correctly attributed under

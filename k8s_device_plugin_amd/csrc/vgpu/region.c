@@ -262,7 +262,8 @@ int vgpu_region_layout_json(char *buf, size_t buflen) {
       "{\"_size\":%lu,\"magic\":%lu,\"version\":%lu,\"init_flag\":%lu,"
       "\"owner_pid\":%lu,\"num_devices\":%lu,\"uuids\":%lu,\"limit\":%lu,"
       "\"sm_limit\":%lu,\"core_tokens\":%lu,\"token_fill_rate\":%lu,"
-      "\"last_refill_ns\":%lu,\"procs\":%lu,\"proc_num\":%lu,"
+      "\"last_refill_ns\":%lu,\"monitor_scale_fp\":%lu,"
+      "\"monitor_scale_ts_ns\":%lu,\"procs\":%lu,\"proc_num\":%lu,"
       "\"utilization_switch\":%lu,\"recent_kernel\":%lu,\"priority\":%lu,"
       "\"oversubscribe\":%lu,\"_proc_slot_size\":%lu,\"_proc_pid\":%lu,"
       "\"_proc_host_pid\":%lu,\"_proc_used\":%lu,\"_proc_monitor_used\":%lu,"
@@ -271,7 +272,8 @@ int vgpu_region_layout_json(char *buf, size_t buflen) {
       (unsigned long)sizeof(vgpu_region_t), OFF(magic), OFF(version),
       OFF(init_flag), OFF(owner_pid), OFF(num_devices), OFF(uuids),
       OFF(limit), OFF(sm_limit), OFF(core_tokens), OFF(token_fill_rate),
-      OFF(last_refill_ns), OFF(procs), OFF(proc_num),
+      OFF(last_refill_ns), OFF(monitor_scale_fp), OFF(monitor_scale_ts_ns),
+      OFF(procs), OFF(proc_num),
       OFF(utilization_switch), OFF(recent_kernel), OFF(priority),
       OFF(oversubscribe), (unsigned long)sizeof(vgpu_proc_slot_t),
       (unsigned long)offsetof(vgpu_proc_slot_t, pid),

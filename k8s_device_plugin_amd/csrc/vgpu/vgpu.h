@@ -27,7 +27,7 @@ extern "C" {
 #endif
 
 #define VGPU_MAGIC 0x4D495655u /* "MIVU" */
-#define VGPU_VERSION 1
+#define VGPU_VERSION 2
 #define VGPU_MAX_DEVICES 16
 #define VGPU_MAX_PROCS 1024
 #define VGPU_UUID_LEN 96
@@ -83,6 +83,13 @@ typedef struct {
   int64_t core_tokens[VGPU_MAX_DEVICES];
   int64_t token_fill_rate[VGPU_MAX_DEVICES]; /* tokens/sec, feedback-adjusted */
   uint64_t last_refill_ns;
+  /* Monitor-arbitrated throttle scale (fixed-point x1e6): the node monitor
+   * sees EVERY container's region, so it writes one per-device scale to all
+   * of them — equal multiplier x entitled CU share = proportional fairness
+   * without per-process utilization attribution (which the kernel cannot
+   * provide for KFD queues).  The limiter uses it while fresh (<2 s). */
+  int64_t monitor_scale_fp[VGPU_MAX_DEVICES];
+  uint64_t monitor_scale_ts_ns;
   vgpu_proc_slot_t procs[VGPU_MAX_PROCS];
   int32_t proc_num;
   int32_t utilization_switch; /* monitor: 1 = enforce CU limit, 0 = free-run */

@@ -14,7 +14,8 @@ Reference behavior (cmd/vGPUmonitor/feedback.go:197-255):
 from __future__ import annotations
 
 import logging
-from typing import Dict, List, Set, Tuple
+import time
+from typing import Callable, Dict, List, Optional, Set, Tuple
 
 from .pathmon import ContainerEntry, PathMonitor
 
@@ -22,14 +23,25 @@ log = logging.getLogger(__name__)
 
 ACTIVITY_THRESHOLD = 0  # recent_kernel above this => active
 
+# node-arbitrated throttle scale bounds (written into every region so all
+# co-located containers share ONE multiplier on their entitled CU share —
+# proportional fairness without per-process attribution)
+SCALE_LO, SCALE_HI = 0.05, 100.0
+BUSY_HIGH, BUSY_LOW = 95, 85
+
 
 class FeedbackLoop:
-    def __init__(self, pathmon: PathMonitor, soft_cores: bool = False):
+    def __init__(self, pathmon: PathMonitor, soft_cores: bool = False,
+                 busy_reader: Optional[Callable[[str], int]] = None):
         """soft_cores=True: enforce the CU limit only while the device is
         contended (the reference's default GPU_CORE_UTILIZATION_POLICY);
-        False (our default): strict isolation, limit always enforced."""
+        False (our default): strict isolation, limit always enforced.
+        busy_reader(uuid) -> device busy percent enables node-arbitrated
+        fair throttling (monitor writes one scale to every region)."""
         self.pathmon = pathmon
         self.soft_cores = soft_cores
+        self.busy_reader = busy_reader
+        self._scale: Dict[str, float] = {}  # device uuid -> multiplier
 
     def observe_once(self) -> None:
         entries = self.pathmon.live_regions()
@@ -78,3 +90,37 @@ class FeedbackLoop:
                 e.region.set_utilization_switch(1 if e.key in contended else 0)
             else:
                 e.region.set_utilization_switch(1)
+
+        if self.busy_reader is not None:
+            self._arbitrate(by_device)
+
+    def _arbitrate(self, by_device) -> None:
+        """One AIMD-style scale per device, written to every region holding
+        it: the limiter multiplies each container's entitled share by this
+        common factor, so shares stay proportional while total device busy
+        converges just under saturation."""
+        now = time.monotonic_ns()
+        for uuid, prio_map in by_device.items():
+            try:
+                busy = self.busy_reader(uuid)
+            except Exception as e:
+                log.warning("busy read for %s failed: %s", uuid, e)
+                continue
+            if busy is None or busy < 0:
+                continue
+            scale = self._scale.get(uuid, 1.0)
+            if busy > BUSY_HIGH:
+                scale *= 0.90
+            elif busy < BUSY_LOW:
+                scale *= 1.10
+            scale = max(SCALE_LO, min(SCALE_HI, scale))
+            self._scale[uuid] = scale
+            for ents in prio_map.values():
+                for e in ents:
+                    try:
+                        snap = e.region.snapshot()
+                        for dev, u in enumerate(snap.uuids):
+                            if u == uuid:
+                                e.region.set_monitor_scale(dev, scale, now)
+                    except Exception as exc:
+                        log.warning("scale write %s failed: %s", e.key, exc)

@@ -73,7 +73,22 @@ def main(argv=None):
 
     gpus = enumerate_gpus()
     pathmon = PathMonitor(args.hook_path)
-    feedback = FeedbackLoop(pathmon, soft_cores=args.soft_cores)
+
+    busy_paths = {g.uuid: f"/sys/class/drm/card{g.drm_card}/device/gpu_busy_percent"
+                  for g in gpus}
+
+    def busy_reader(uuid: str):
+        path = busy_paths.get(uuid)
+        if path is None:
+            return -1
+        try:
+            with open(path) as f:
+                return int(f.read().strip())
+        except (OSError, ValueError):
+            return -1
+
+    feedback = FeedbackLoop(pathmon, soft_cores=args.soft_cores,
+                            busy_reader=busy_reader if gpus else None)
     collector = MonitorCollector(pathmon, gpus)
     serve_metrics(collector, args.metrics_port)
     if args.grpc_bind:

@@ -113,6 +113,20 @@ class SharedRegion:
                + self.layout["_proc_host_pid"])
         self._w_i32(off, host_pid)
 
+    def set_monitor_scale(self, dev: int, scale: float, now_ns: int) -> None:
+        """Write the node-arbitrated throttle scale (fixed-point x1e6) and
+        freshness timestamp; the limiter uses it while < 2 s old."""
+        struct.pack_into("<q", self._mm,
+                         self.layout["monitor_scale_fp"] + 8 * dev,
+                         int(scale * 1e6))
+        struct.pack_into("<Q", self._mm,
+                         self.layout["monitor_scale_ts_ns"], now_ns)
+
+    def get_monitor_scale(self, dev: int) -> float:
+        v = struct.unpack_from(
+            "<q", self._mm, self.layout["monitor_scale_fp"] + 8 * dev)[0]
+        return v / 1e6
+
     # -- snapshot ------------------------------------------------------------
     def snapshot(self) -> RegionSnapshot:
         L = self.layout

@@ -236,3 +236,72 @@ class TestNodeRPC:
                 server.stop(0)
             proc.kill()
             proc.wait()
+
+
+class TestArbitratedThrottle:
+    """Node-arbitrated fair throttling: the monitor writes ONE per-device
+    scale into every co-located region; the C limiter's refill uses it
+    while fresh (<2 s), overriding its local feedback loop.  Proportional
+    fairness without per-process utilization attribution (which the kernel
+    cannot provide for KFD queues — profiles/r01_summary.md)."""
+
+    def _start_consumer(self, cache, cu=10):
+        return subprocess.Popen(
+            [str(CONSUMER), "launch", "1", "1", "sleep", "30000"],
+            env=consumer_env(cache, extra={"VGPU_DEVICE_CU_LIMIT": str(cu)}),
+            stdout=subprocess.PIPE, text=True)
+
+    def test_scale_written_equally_and_used_by_limiter(self, tmp_path):
+        hook = tmp_path / "hook"
+        caches = []
+        procs = []
+        for i in range(2):
+            d = hook / "containers" / f"pod-arb-{i}_main"
+            d.mkdir(parents=True)
+            cache = d / "r.cache"
+            caches.append(cache)
+            procs.append(self._start_consumer(cache))
+        try:
+            for p in procs:
+                json.loads(p.stdout.readline())  # first launch done
+
+            pathmon = PathMonitor(str(hook))
+            pathmon.scan({"pod-arb-0", "pod-arb-1"})
+            # both fake regions carry uuid GPU-test-0 (consumer_env)
+            fb = FeedbackLoop(pathmon, busy_reader=lambda uuid: 99)
+            # mark both active so they appear in by_device
+            for e in pathmon.live_regions():
+                e.region.set_recent_kernel(5)
+            for _ in range(5):
+                fb.observe_once()   # busy 99 > 95 -> scale shrinks each tick
+            scales = [SharedRegion(str(c)).get_monitor_scale(0) for c in caches]
+            assert scales[0] == scales[1]           # same multiplier for all
+            assert scales[0] < 1.0                   # shrunk under saturation
+            # C side: a fresh scale drives token_fill_rate = RATE_FULL *
+            # lim/100 * scale on the next refill ticks
+            import struct as _struct
+            import time as _time
+
+            region = SharedRegion(str(caches[0]))
+            deadline = _time.time() + 10
+            expect = 4_000_000 * 0.10 * scales[0]
+            ok = False
+            while _time.time() < deadline:
+                fb.observe_once()  # keep the ts fresh
+                rate = _struct.unpack_from(
+                    "<q", region._mm, region.layout["token_fill_rate"])[0]
+                if rate and abs(rate - expect) / expect < 0.25:
+                    ok = True
+                    break
+                _time.sleep(0.2)
+            assert ok, f"limiter did not adopt monitor scale (rate={rate}, expect={expect})"
+
+            # scale recovers when the device is idle
+            fb2 = FeedbackLoop(pathmon, busy_reader=lambda uuid: 10)
+            fb2._scale.update(fb._scale)
+            fb2.observe_once()
+            assert fb2._scale["GPU-test-0"] > fb._scale["GPU-test-0"] * 0.99
+        finally:
+            for p in procs:
+                p.kill()
+                p.wait()
