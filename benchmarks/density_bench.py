@@ -67,6 +67,10 @@ def main():
                    help="CU percent per pod (default 100/pods)")
     p.add_argument("--no-cu-mask", action="store_true",
                    help="skip the hard HSA_CU_MASK partition (soft limiter only)")
+    p.add_argument("--arbitrate", action="store_true",
+                   help="run the monitor's fair-throttle arbitration loop "
+                        "over the pods' regions (as the vGPU monitor does "
+                        "in production)")
     p.add_argument("--prewarm", action="store_true",
                    help="run one unmasked pass first to populate the MIOpen "
                         "find-db (keyed by conv config, not CU mask) so the "
@@ -135,6 +139,53 @@ def main():
              "--seconds", str(args.seconds)],
             env=env, stdin=subprocess.PIPE, stdout=subprocess.PIPE, text=True))
 
+    arb_stop = None
+    if args.arbitrate:
+        import glob
+        import threading
+
+        from k8s_device_plugin_amd.monitor.region import SharedRegion
+
+        def find_busy_path():
+            for path in sorted(glob.glob(
+                    "/sys/class/drm/card*/device/gpu_busy_percent")):
+                return path
+            return None
+
+        busy_path = find_busy_path()
+        arb_stop = threading.Event()
+
+        def arbitrate():
+            regions = []
+            for c in caches:
+                try:
+                    r = SharedRegion(c)
+                    if r.valid:
+                        regions.append(r)
+                except (OSError, ValueError):
+                    pass
+            scale = 1.0
+            while not arb_stop.is_set():
+                busy = -1
+                if busy_path:
+                    try:
+                        busy = int(open(busy_path).read().strip())
+                    except (OSError, ValueError):
+                        pass
+                if busy > 95:
+                    scale = max(0.05, scale * 0.90)
+                elif 0 <= busy < 85:
+                    scale = min(100.0, scale * 1.10)
+                now = time.monotonic_ns()
+                for r in regions:
+                    try:
+                        r.set_monitor_scale(0, scale, now)
+                    except (OSError, ValueError):
+                        pass
+                arb_stop.wait(0.25)
+
+        threading.Thread(target=arbitrate, daemon=True).start()
+
     # barrier: wait until every pod is warm, then release together
     for pr in procs:
         assert pr.stdout.readline().strip() == "READY"
@@ -150,6 +201,8 @@ def main():
                 break
         pr.wait(timeout=120)
         results.append(json.loads(line))
+    if arb_stop is not None:
+        arb_stop.set()
     for c in caches:
         os.unlink(c)
 
@@ -164,6 +217,7 @@ def main():
         "quota_pct": quota_pct,
         "cu_pct": cu_pct,
         "hard_cu_mask": not args.no_cu_mask,
+        "arbitrated": bool(args.arbitrate),
         "case": args.case,
         "seconds": args.seconds,
     }))
