@@ -156,16 +156,20 @@ def main():
         arb_stop = threading.Event()
 
         def arbitrate():
-            regions = []
-            for c in caches:
-                try:
-                    r = SharedRegion(c)
-                    if r.valid:
-                        regions.append(r)
-                except (OSError, ValueError):
-                    pass
+            regions = {}
             scale = 1.0
             while not arb_stop.is_set():
+                # late-attach: the workers create/initialize their regions
+                # after launch
+                for c in caches:
+                    if c in regions:
+                        continue
+                    try:
+                        r = SharedRegion(c)
+                        if r.valid:
+                            regions[c] = r
+                    except (OSError, ValueError):
+                        pass
                 busy = -1
                 if busy_path:
                     try:
@@ -177,7 +181,7 @@ def main():
                 elif 0 <= busy < 85:
                     scale = min(100.0, scale * 1.10)
                 now = time.monotonic_ns()
-                for r in regions:
+                for r in regions.values():
                     try:
                         r.set_monitor_scale(0, scale, now)
                     except (OSError, ValueError):
