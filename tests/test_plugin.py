@@ -538,3 +538,51 @@ class TestMultiContainerAllocate:
         # both consumed -> success + lock released
         assert client.get_pod("p2").annotations[BIND_PHASE_ANNO] == BIND_PHASE_SUCCESS
         assert NODE_LOCK_ANNO not in client.get_node("node1").annotations
+
+
+class TestNodeConfigOverride:
+    """Per-node JSON ConfigMap precedence: node JSON > CLI > defaults
+    (reference vgpucfg.go:81-107)."""
+
+    def test_matching_node_overridden(self, tmp_path):
+        from k8s_device_plugin_amd.plugin.config import (
+            PluginConfig,
+            apply_node_config,
+        )
+
+        cfgfile = tmp_path / "config.json"
+        cfgfile.write_text(json.dumps({"nodeconfig": [
+            {"name": "other", "devicesplitcount": 2},
+            {"name": "n1", "devicesplitcount": 4,
+             "devicememoryscaling": 1.5, "devicecorescaling": 0.5},
+        ]}))
+        cfg = PluginConfig(node_name="n1", device_split_count=10,
+                           config_file=str(cfgfile))
+        out = apply_node_config(cfg)
+        assert out.device_split_count == 4
+        assert out.device_memory_scaling == 1.5
+        assert out.device_cores_scaling == 0.5
+
+    def test_non_matching_node_untouched(self, tmp_path):
+        from k8s_device_plugin_amd.plugin.config import (
+            PluginConfig,
+            apply_node_config,
+        )
+
+        cfgfile = tmp_path / "config.json"
+        cfgfile.write_text(json.dumps({"nodeconfig": [
+            {"name": "other", "devicesplitcount": 2}]}))
+        cfg = PluginConfig(node_name="n1", device_split_count=10,
+                           config_file=str(cfgfile))
+        assert apply_node_config(cfg).device_split_count == 10
+
+    def test_bad_json_ignored(self, tmp_path):
+        from k8s_device_plugin_amd.plugin.config import (
+            PluginConfig,
+            apply_node_config,
+        )
+
+        cfgfile = tmp_path / "config.json"
+        cfgfile.write_text("{nope")
+        cfg = PluginConfig(node_name="n1", config_file=str(cfgfile))
+        assert apply_node_config(cfg).device_split_count == 10

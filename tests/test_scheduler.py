@@ -372,3 +372,51 @@ class TestPodCacheEviction:
         usage, _ = sched.get_nodes_usage(["n1"])
         assert usage["n1"].devices[0].usedmem == 0
         assert usage["n1"].devices[0].used == 0
+
+
+class TestSchedulerMetrics:
+    """All 10 reference metric families (cmd/scheduler/metrics.go:49-190)
+    render with data."""
+
+    def test_families_present(self):
+        from k8s_device_plugin_amd.device.amd import REGISTER_ANNO, HANDSHAKE_ANNO
+        from k8s_device_plugin_amd.scheduler.core import Scheduler
+        from k8s_device_plugin_amd.scheduler.metrics import metrics_text
+        from k8s_device_plugin_amd.utils.codec import (
+            encode_node_devices,
+            encode_pod_single_device,
+        )
+        from k8s_device_plugin_amd.utils.kubeclient import FakeKubeClient
+        from k8s_device_plugin_amd.utils.types import (
+            ASSIGNED_NODE_ANNO,
+            SUPPORT_DEVICES,
+            ContainerDevice,
+            DeviceInfo,
+            NodeInfo,
+            PodInfo,
+        )
+
+        client = FakeKubeClient()
+        client.add_node(NodeInfo(name="n1", annotations={
+            HANDSHAKE_ANNO: "Reported 2026-01-01 00:00:00",
+            REGISTER_ANNO: encode_node_devices([DeviceInfo(
+                id="GPU-m", count=10, devmem=294912, devcore=100,
+                type="AMD-Instinct-MI355X", numa=0, health=True, index=0)]),
+        }))
+        sched = Scheduler(client)
+        sched.register_from_node_annotations_once()
+        client.add_pod(PodInfo(name="p", uid="uid-m", node_name="n1", annotations={
+            ASSIGNED_NODE_ANNO: "n1",
+            SUPPORT_DEVICES["AMD"]: encode_pod_single_device(
+                [[ContainerDevice(uuid="GPU-m", type="AMD", usedmem=1024,
+                                  usedcores=25)]]),
+        }))
+        sched.rebuild_pod_cache()
+        sched.get_nodes_usage(["n1"])
+        text = metrics_text(sched).decode()
+        for family in ["GPUDeviceMemoryLimit", "GPUDeviceCoreLimit",
+                       "GPUDeviceMemoryAllocated", "GPUDeviceSharedNum",
+                       "GPUDeviceCoreAllocated", "nodeGPUOverview",
+                       "nodeGPUMemoryPercentage", "vGPUPodsDeviceAllocated",
+                       "vGPUMemoryPercentage", "vGPUCorePercentage"]:
+            assert family in text, f"missing metric family {family}"
