@@ -488,3 +488,21 @@ hipError_t hipGraphLaunch(void *graphExec, void *stream) {
     vgpu_limiter_gate(vgpu_current_device(), 2048);
   return real(graphExec, stream);
 }
+
+/* hipExtLaunchKernelGGL lowers to hipExtLaunchKernel (events + flags
+ * variant); same gate as hipLaunchKernel. */
+hipError_t hipExtLaunchKernel(const void *f, vdim3 grid, vdim3 block,
+                              void **args, size_t shared, void *stream,
+                              void *startEvent, void *stopEvent, int flags) {
+  typedef hipError_t (*fn)(const void *, vdim3, vdim3, void **, size_t,
+                           void *, void *, void *, int);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipExtLaunchKernel");
+  if (!real) return hipErrorInvalidValue;
+  vgpu_ensure_initialized();
+  if (!vgpu_control_disabled())
+    vgpu_limiter_gate(vgpu_current_device(),
+                      (uint64_t)grid.x * grid.y * grid.z);
+  return real(f, grid, block, args, shared, stream, startEvent, stopEvent,
+              flags);
+}

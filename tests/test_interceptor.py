@@ -226,3 +226,25 @@ class TestRegionABI:
         lib.vgpu_region_layout_json(buf, 4096)
         layout = json.loads(buf.value.decode())
         assert cache.stat().st_size == layout["_size"]
+
+
+class TestActiveOOMKiller:
+    def test_over_quota_exits_137(self, tmp_path):
+        """ACTIVE_OOM_KILLER: an over-quota allocation aborts the process
+        (reference active_oom_killer, SURVEY.md §2.6 'Memory cap')."""
+        env = dict(os.environ)
+        env.update({
+            "LD_LIBRARY_PATH": str(FAKEDIR),
+            "LD_PRELOAD": str(LIBVGPU),
+            "VGPU_DEVICE_MEMORY_SHARED_CACHE": str(tmp_path / "r.cache"),
+            "VGPU_REAL_HIP_PATH": str(FAKEDIR / "libamdhip64.so"),
+            "VGPU_DEVICE_MEMORY_LIMIT": "1000m",
+            "ACTIVE_OOM_KILLER": "1",
+        })
+        out = subprocess.run(
+            [str(CONSUMER), "alloc", str(600 * MIB), "alloc", str(600 * MIB),
+             "meminfo"],
+            env=env, capture_output=True, text=True, timeout=120)
+        assert out.returncode == 137          # killed at the second alloc
+        lines = [json.loads(l) for l in out.stdout.splitlines()]
+        assert len(lines) == 1 and lines[0]["err"] == 0
