@@ -69,6 +69,7 @@ class Scheduler:
         self.overview_status: Dict[str, NodeUsage] = {}
         self.cached_status: Dict[str, NodeUsage] = {}
         self._stop = threading.Event()
+        self._filter_lock = threading.Lock()
         self._lock = threading.RLock()
 
     # ------------------------------------------------------------------
@@ -239,6 +240,13 @@ class Scheduler:
     # Filter / Bind
     # ------------------------------------------------------------------
     def filter(self, pod: PodInfo, node_names: List[str]) -> FilterResult:
+        # kube-scheduler runs one scheduling cycle at a time, but the HTTP
+        # server is threaded: serialize snapshot->score->commit so a custom
+        # client can never over-commit a device
+        with self._filter_lock:
+            return self._filter_locked(pod, node_names)
+
+    def _filter_locked(self, pod: PodInfo, node_names: List[str]) -> FilterResult:
         nums = pod_device_requests(pod)
         total = sum(int(k.nums) for n in nums for k in n.values())
         if total == 0:

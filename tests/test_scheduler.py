@@ -420,3 +420,56 @@ class TestSchedulerMetrics:
                        "nodeGPUMemoryPercentage", "vGPUPodsDeviceAllocated",
                        "vGPUMemoryPercentage", "vGPUCorePercentage"]:
             assert family in text, f"missing metric family {family}"
+
+
+class TestConcurrentFilter:
+    """Parallel /filter calls must keep accounting consistent (the
+    reference guards its managers with mutexes, nodes.go:50-53)."""
+
+    def test_parallel_filters_never_overcommit(self):
+        import threading
+
+        from k8s_device_plugin_amd.device.amd import REGISTER_ANNO, HANDSHAKE_ANNO
+        from k8s_device_plugin_amd.scheduler.core import Scheduler
+        from k8s_device_plugin_amd.utils.codec import encode_node_devices
+        from k8s_device_plugin_amd.utils.kubeclient import FakeKubeClient
+        from k8s_device_plugin_amd.utils.types import (
+            ContainerSpec,
+            DeviceInfo,
+            NodeInfo,
+            PodInfo,
+        )
+
+        client = FakeKubeClient()
+        client.add_node(NodeInfo(name="n1", annotations={
+            HANDSHAKE_ANNO: "Reported 2026-01-01 00:00:00",
+            REGISTER_ANNO: encode_node_devices([DeviceInfo(
+                id="GPU-c", count=4, devmem=294912, devcore=100,
+                type="AMD-Instinct-MI355X", numa=0, health=True, index=0)]),
+        }))
+        sched = Scheduler(client)
+        sched.register_from_node_annotations_once()
+
+        results = []
+
+        def one(i):
+            pod = PodInfo(
+                name=f"p{i}", uid=f"uid-{i}",
+                containers=[ContainerSpec(
+                    name="c", limits={"amd.com/gpu": 1,
+                                      "amd.com/gpumem": 100000})],
+            )
+            client.add_pod(pod)
+            results.append(sched.filter(pod, ["n1"]))
+
+        threads = [threading.Thread(target=one, args=(i,)) for i in range(6)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        # 294912 MiB total, 100000 each -> exactly 2 fit (filter is
+        # serialized snapshot->commit)
+        placed = [r for r in results if r.node_names]
+        assert len(placed) == 2
+        usage, _ = sched.get_nodes_usage(["n1"])
+        assert usage["n1"].devices[0].usedmem <= 294912
