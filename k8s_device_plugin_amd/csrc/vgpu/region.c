@@ -115,6 +115,18 @@ static void region_register_proc(vgpu_region_t *r) {
     s->pid = me;
     s->status = 1;
     g_proc_slot = free_slot;
+    /* charge the configured per-process context overhead on every device
+     * this container sees, so quota math covers runtime reservations */
+    const char *ctx = getenv(ENV_CONTEXT_OVERHEAD);
+    if (ctx && *ctx) {
+      uint64_t bytes = parse_size(ctx);
+      uint64_t n = r->num_devices ? r->num_devices : 1;
+      for (uint64_t d = 0; d < n && d < VGPU_MAX_DEVICES; d++) {
+        s->used[d].context_size = bytes;
+        s->used[d].total = s->used[d].context_size +
+                           s->used[d].module_size + s->used[d].buffer_size;
+      }
+    }
     int cnt = 0;
     for (int i = 0; i < VGPU_MAX_PROCS; i++)
       if (r->procs[i].pid && vgpu_proc_alive(r->procs[i].pid)) cnt++;

@@ -47,6 +47,10 @@ extern "C" {
 #define ENV_REAL_HSA "VGPU_REAL_HSA_PATH"
 #define ENV_REAL_AMDSMI "VGPU_REAL_AMDSMI_PATH"
 #define ENV_DEVICE_UUIDS "VGPU_DEVICE_UUIDS"          /* comma list, monitor correlation */
+/* per-process runtime/context overhead charged at region registration
+ * (the reference charges CUDA context size the same way, SURVEY §2.6
+ * "context/module split"); size string, e.g. "512m"; default 0 */
+#define ENV_CONTEXT_OVERHEAD "VGPU_CONTEXT_OVERHEAD"
 
 /* memory accounting split, per device per process (reference ABI:
  * cmd/vGPUmonitor/cudevshr.go:15-58 deviceMemory) */

@@ -248,3 +248,52 @@ class TestActiveOOMKiller:
         assert out.returncode == 137          # killed at the second alloc
         lines = [json.loads(l) for l in out.stdout.splitlines()]
         assert len(lines) == 1 and lines[0]["err"] == 0
+
+
+class TestContextOverhead:
+    def test_context_charge_counts_against_quota(self, tmp_path):
+        """VGPU_CONTEXT_OVERHEAD charges the runtime's per-process
+        reservation into the context bucket (reference context_size,
+        SURVEY.md §2.6) — visible in meminfo and in the monitor's
+        context/module/data breakdown."""
+        res = run_consumer(
+            ["meminfo", "alloc", str(700 * MIB), "meminfo"],
+            tmp_path / "r.cache", mem_limit="1000m",
+            extra_env={"VGPU_CONTEXT_OVERHEAD": "200m"},
+        )
+        assert res[0]["free"] == 800 * MIB      # context charged up front
+        assert res[1]["err"] == 0               # 200 + 700 <= 1000
+        assert res[2]["free"] == 100 * MIB
+        # and an alloc that would fit without the context charge fails
+        res = run_consumer(
+            ["alloc", str(900 * MIB)],
+            tmp_path / "r2.cache", mem_limit="1000m",
+            extra_env={"VGPU_CONTEXT_OVERHEAD": "200m"},
+        )
+        assert res[0]["err"] == 2
+
+    def test_monitor_sees_context_split(self, tmp_path):
+        import subprocess as sp
+
+        from k8s_device_plugin_amd.monitor.region import SharedRegion
+
+        cache = tmp_path / "r.cache"
+        proc = sp.Popen(
+            [str(CONSUMER), "alloc", str(100 * MIB), "sleep", "8000"],
+            env={**os.environ,
+                 "LD_LIBRARY_PATH": str(FAKEDIR),
+                 "LD_PRELOAD": str(LIBVGPU),
+                 "VGPU_DEVICE_MEMORY_SHARED_CACHE": str(cache),
+                 "VGPU_REAL_HIP_PATH": str(FAKEDIR / "libamdhip64.so"),
+                 "VGPU_DEVICE_MEMORY_LIMIT": "1000m",
+                 "VGPU_CONTEXT_OVERHEAD": "150m"},
+            stdout=sp.PIPE, text=True)
+        try:
+            json.loads(proc.stdout.readline())
+            region = SharedRegion(str(cache))
+            snap = region.snapshot()
+            p0 = snap.procs[0]
+            assert p0.used_bytes[0] == 250 * MIB  # 150 ctx + 100 buffer
+        finally:
+            proc.kill()
+            proc.wait()
