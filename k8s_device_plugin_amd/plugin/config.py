@@ -34,6 +34,9 @@ class PluginConfig:
     config_file: str = "/config/config.json"
     register_interval_s: float = 30.0
     health_interval_s: float = 5.0
+    # per-process runtime reservation charged against the quota inside the
+    # container (VGPU_CONTEXT_OVERHEAD); 0 = off
+    context_overhead_mb: int = 0
     # "envvar" (ROCR_VISIBLE_DEVICES + DeviceSpecs) or "cdi-annotations"
     # (additionally name CDI devices in the Allocate response annotations;
     # reference --device-list-strategy, main.go:61-70)
@@ -56,6 +59,7 @@ def parse_args(argv: Optional[List[str]] = None) -> PluginConfig:
     p.add_argument("--config-file", default=c.config_file)
     p.add_argument("--device-list-strategy", default=c.device_list_strategy,
                    choices=["envvar", "cdi-annotations"])
+    p.add_argument("--context-overhead-mb", type=int, default=c.context_overhead_mb)
     p.add_argument("--cdi-spec-dir", default=c.cdi_spec_dir)
     a = p.parse_args(argv)
     cfg = PluginConfig(
@@ -71,6 +75,7 @@ def parse_args(argv: Optional[List[str]] = None) -> PluginConfig:
         config_file=a.config_file,
         device_list_strategy=a.device_list_strategy,
         cdi_spec_dir=a.cdi_spec_dir,
+        context_overhead_mb=a.context_overhead_mb,
     )
     return apply_node_config(cfg)
 

@@ -214,6 +214,8 @@ class VGPUDevicePlugin:
             resp.envs["VGPU_SYSFS_CARDS"] = ",".join(cards)
         if rsmi_indices:
             resp.envs["VGPU_RSMI_INDICES"] = ",".join(rsmi_indices)
+        if cfg.context_overhead_mb > 0:
+            resp.envs["VGPU_CONTEXT_OVERHEAD"] = f"{cfg.context_overhead_mb}m"
         if cfg.device_memory_scaling > 1:
             resp.envs["VGPU_OVERSUBSCRIBE"] = "true"
             resp.envs["HSA_XNACK"] = "1"
