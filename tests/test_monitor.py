@@ -305,3 +305,40 @@ class TestArbitratedThrottle:
             for p in procs:
                 p.kill()
                 p.wait()
+
+
+class TestVgpuTop:
+    def test_rows_and_render(self, tmp_path):
+        from k8s_device_plugin_amd.tools import vgpu_top
+
+        hook = tmp_path / "hook"
+        d = hook / "containers" / "pod-top-1_main"
+        d.mkdir(parents=True)
+        cache = d / "r.cache"
+        proc = subprocess.Popen(
+            [str(CONSUMER), "alloc", str(256 * MIB), "sleep", "8000"],
+            env=consumer_env(cache), stdout=subprocess.PIPE, text=True)
+        try:
+            json.loads(proc.stdout.readline())
+            pm = PathMonitor(str(hook))
+            pm.scan({"pod-top-1"})
+            rs = vgpu_top.rows(pm)
+            assert len(rs) == 1
+            r = rs[0]
+            assert r["pod"] == "pod-top-1" and r["ctr"] == "main"
+            assert abs(r["used_gib"] - 0.25) < 0.01
+            assert abs(r["limit_gib"] - 1000 / 1024) < 0.01
+            text = vgpu_top.render(rs)
+            assert "pod-top-1" in text and "LIMIT" in text
+            assert vgpu_top.render([])  # empty table renders too
+        finally:
+            proc.kill()
+            proc.wait()
+
+    def test_cli_main_once(self, tmp_path, capsys):
+        from k8s_device_plugin_amd.tools import vgpu_top
+
+        hook = tmp_path / "hook"
+        (hook / "containers").mkdir(parents=True)
+        assert vgpu_top.main(["--hook-path", str(hook)]) == 0
+        assert "no live vGPU containers" in capsys.readouterr().out
