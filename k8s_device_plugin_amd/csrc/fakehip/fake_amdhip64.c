@@ -65,7 +65,9 @@ hipError_t hipSetDevice(int dev) {
 static hipError_t fake_alloc(void **ptr, size_t size, int managed) {
   if (!ptr) return hipErrorInvalidValue;
   pthread_mutex_lock(&g_mu);
-  if (g_alloc[t_current] + size > total_mem()) {
+  /* managed (XNACK/UVM) allocations may exceed device memory — they page
+   * to host DRAM; only plain device allocs are bounded by HBM */
+  if (!managed && g_alloc[t_current] + size > total_mem()) {
     pthread_mutex_unlock(&g_mu);
     return hipErrorOutOfMemory;
   }

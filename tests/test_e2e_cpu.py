@@ -152,3 +152,81 @@ def test_pod_lifecycle_end_to_end(tmp_path):
     assert "vGPU_device_memory_limit_in_bytes" in mtext
 
     plugin.stop()
+
+
+def test_oversubscription_end_to_end(tmp_path):
+    """BASELINE config 5 acceptance on CPU: deviceMemoryScaling=1.4 -> the
+    node advertises ~400 GB on a 288 GB card, a pod requests beyond
+    physical, Allocate injects VGPU_OVERSUBSCRIBE, and the container's
+    device allocations become managed (fake-runtime managed counter)."""
+    client = FakeKubeClient()
+    client.add_node(NodeInfo(name="n1"))
+    topo, pci = make_kfd_tree(tmp_path, n_gpus=1)
+    rm = ResourceManager(enumerate_gpus(str(topo), str(pci)), split_count=4,
+                         memory_scaling=1.4)
+    cfg = PluginConfig(
+        node_name="n1",
+        hook_path=str(tmp_path / "hook"),
+        plugin_socket_dir=str(tmp_path),
+        kubelet_socket=str(tmp_path / "kubelet.sock"),
+        device_memory_scaling=1.4,
+    )
+    plugin = VGPUDevicePlugin(cfg, rm, client)
+    plugin.serve()
+    register_once(client, "n1", rm)
+    sched = Scheduler(client)
+    sched.register_from_node_annotations_once()
+
+    # the advertised capacity is scaled past physical
+    adv = rm.api_devices()[0].devmem
+    assert adv > 288 * 1024  # MiB
+
+    pod_obj = {
+        "kind": "Pod",
+        "metadata": {"name": "big-1", "namespace": "default", "uid": "uid-big-1"},
+        "spec": {"containers": [{
+            "name": "main",
+            "resources": {"limits": {
+                "amd.com/gpu": "1",
+                "amd.com/gpumem": "409600",   # 400 GiB on a 288 GiB card
+            }},
+        }]},
+    }
+    pod = PodInfo.from_k8s(pod_obj)
+    client.add_pod(pod)
+    fr = sched.filter(pod, ["n1"])
+    assert fr.node_names == ["n1"], fr
+    assert not sched.bind("big-1", "default", "n1").error
+
+    with grpc.insecure_channel(f"unix://{plugin.socket_path}") as ch:
+        stub = dp.DevicePluginClient(ch)
+        req = dp.AllocateRequest()
+        req.container_requests.add(devicesIDs=[f"{rm.gpus[0].uuid}-0"])
+        resp = stub.Allocate(req)
+    envs = dict(resp.container_responses[0].envs)
+    assert envs["VGPU_OVERSUBSCRIBE"] == "true"
+    assert envs["HSA_XNACK"] == "1"
+    assert envs["VGPU_DEVICE_MEMORY_LIMIT_0"] == "409600m"
+
+    # inside the "container": allocations route to managed memory
+    ctr_dir = Path(cfg.hook_path) / "vgpu" / "containers" / "uid-big-1_main"
+    run_env = dict(os.environ)
+    run_env.update({
+        "LD_LIBRARY_PATH": str(CSRC / "fakehip"),
+        "LD_PRELOAD": str(CSRC / "libvgpu-hip.so"),
+        "VGPU_REAL_HIP_PATH": str(CSRC / "fakehip" / "libamdhip64.so"),
+        "VGPU_DEVICE_MEMORY_LIMIT": envs["VGPU_DEVICE_MEMORY_LIMIT_0"],
+        "VGPU_OVERSUBSCRIBE": envs["VGPU_OVERSUBSCRIBE"],
+        "VGPU_DEVICE_MEMORY_SHARED_CACHE": str(ctr_dir / "region.cache"),
+        "FAKE_HIP_TOTAL_MEM": str(288 << 30),
+    })
+    out = subprocess.run(
+        [str(CSRC / "test" / "hip_consumer"),
+         "meminfo", "alloc", str(300 * 1024 * MIB), "stats"],
+        env=run_env, capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    lines = [json.loads(l) for l in out.stdout.splitlines()]
+    assert lines[0]["total"] == 409600 * MIB    # virtual capacity visible
+    assert lines[1]["err"] == 0                 # 300 GiB > physical succeeds
+    assert lines[2]["managed"] >= 1             # went through hipMallocManaged
+    plugin.stop()
