@@ -20,6 +20,8 @@
 #include <stdio.h>
 #include <stdlib.h>
 #include <string.h>
+#include <sys/wait.h>
+#include <unistd.h>
 #include <time.h>
 
 typedef int hipError_t;
@@ -97,6 +99,30 @@ int main(int argc, char **argv) {
       if (mf) managed = (long)mf();
       printf("{\"cmd\":\"stats\",\"launches\":%ld,\"managed\":%ld}\n",
              launches, managed);
+    } else if (strcmp(cmd, "forkhold") == 0 && i + 2 < argc) {
+      /* fork(): the child's allocations must account to ITS OWN proc slot
+       * (atfork re-registration) and vanish when it exits */
+      size_t n = strtoull(argv[++i], NULL, 10);
+      long ms = atol(argv[++i]);
+      pid_t pid = fork();
+      if (pid == 0) {
+        void *p = NULL;
+        hipError_t e = hipMalloc(&p, n);
+        struct timespec ts;
+        ts.tv_sec = ms / 1000;
+        ts.tv_nsec = (ms % 1000) * 1000000L;
+        nanosleep(&ts, NULL);
+        _exit(e == 0 ? 0 : 42);
+      }
+      struct timespec w = {0, 300 * 1000000L};
+      nanosleep(&w, NULL);
+      size_t f = 0, t = 0;
+      hipMemGetInfo(&f, &t);
+      printf("{\"cmd\":\"forkhold\",\"free\":%zu,\"total\":%zu}\n", f, t);
+      fflush(stdout);
+      int st = 0;
+      waitpid(pid, &st, 0);
+      printf("{\"cmd\":\"forkdone\",\"status\":%d}\n", WEXITSTATUS(st));
     } else if (strcmp(cmd, "sleep") == 0 && i + 1 < argc) {
       struct timespec ts;
       long ms = atol(argv[++i]);

@@ -137,6 +137,8 @@ static void region_register_proc(vgpu_region_t *r) {
   vgpu_region_unlock(r);
 }
 
+static void atfork_child(void);
+
 static void region_open(void) {
   const char *path = getenv(ENV_SHARED_CACHE);
   char fallback[PATH_MAX];
@@ -182,6 +184,18 @@ static void region_open(void) {
   close(fd); /* mapping persists */
   g_region = r;
   region_register_proc(r);
+  pthread_atfork(NULL, NULL, atfork_child);
+}
+
+/* fork() support: the child inherits the mapping but NOT the proc slot —
+ * its pid differs, so its allocations must go to its own slot (PyTorch
+ * dataloader workers etc.).  Registered once via pthread_atfork
+ * (forward-declared above region_open). */
+static void atfork_child(void) {
+  if (g_region) {
+    g_proc_slot = -1;
+    region_register_proc(g_region);
+  }
 }
 
 vgpu_region_t *vgpu_region_get(void) {

@@ -297,3 +297,19 @@ class TestContextOverhead:
         finally:
             proc.kill()
             proc.wait()
+
+
+class TestForkSafety:
+    def test_forked_child_accounts_separately(self, tmp_path):
+        """A forked child re-registers its own proc slot (pthread_atfork):
+        its allocation counts while it lives and is pruned when it exits."""
+        res = run_consumer(
+            ["forkhold", str(300 * MIB), "800", "meminfo"],
+            tmp_path / "r.cache", mem_limit="1000m",
+        )
+        by_cmd = {r["cmd"]: r for r in res}
+        # while the child sleeps, its 300M shows in the shared accounting
+        assert by_cmd["forkhold"]["free"] == 700 * MIB
+        assert by_cmd["forkdone"]["status"] == 0
+        # after the child exits, liveness pruning returns the memory
+        assert by_cmd["meminfo"]["free"] == 1000 * MIB
