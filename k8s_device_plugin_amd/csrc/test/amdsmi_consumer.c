@@ -13,6 +13,8 @@ typedef struct {
 } amdsmi_vram_usage_t;
 
 extern amdsmi_status_t amdsmi_init(uint64_t);
+extern amdsmi_status_t amdsmi_get_socket_handles(uint32_t *,
+                                                 amdsmi_socket_handle *);
 extern amdsmi_status_t amdsmi_get_processor_handles(amdsmi_socket_handle,
                                                     uint32_t *,
                                                     amdsmi_processor_handle *);
@@ -24,10 +26,17 @@ extern amdsmi_status_t amdsmi_get_gpu_vram_usage(amdsmi_processor_handle,
                                                  amdsmi_vram_usage_t *);
 
 int main(void) {
-  amdsmi_init(0);
+  /* AMDSMI_INIT_AMD_GPUS = 1 (real lib requires a flag; fake ignores) */
+  amdsmi_init(1);
+  amdsmi_socket_handle sockets[8];
+  uint32_t nsock = 8;
+  if (amdsmi_get_socket_handles(&nsock, sockets) != 0 || nsock == 0) {
+    printf("enumerate failed\n");
+    return 1;
+  }
   amdsmi_processor_handle hs[16];
   uint32_t n = 16;
-  if (amdsmi_get_processor_handles(NULL, &n, hs) != 0) {
+  if (amdsmi_get_processor_handles(sockets[0], &n, hs) != 0) {
     printf("enumerate failed\n");
     return 1;
   }
