@@ -26,8 +26,8 @@ extern amdsmi_status_t amdsmi_get_gpu_vram_usage(amdsmi_processor_handle,
                                                  amdsmi_vram_usage_t *);
 
 int main(void) {
-  /* AMDSMI_INIT_AMD_GPUS = 1 (real lib requires a flag; fake ignores) */
-  amdsmi_init(1);
+  /* AMDSMI_INIT_AMD_GPUS = 1<<1 (amdsmi.h:51; the fake ignores it) */
+  amdsmi_init(1 << 1);
   amdsmi_socket_handle sockets[8];
   uint32_t nsock = 8;
   if (amdsmi_get_socket_handles(&nsock, sockets) != 0 || nsock == 0) {
