@@ -186,7 +186,10 @@ void *dlsym(void *handle, const char *symbol) {
   dlsym_fn real = real_dlsym();
   if (!real) return NULL;
   void *p = real(handle, symbol);
-  if (!p && symbol && g_self_handle && handle == g_self_handle) {
+  /* volatile copy: glibc marks symbol __nonnull, but defensive NULL
+   * handling matters for an interposer (-Wnonnull-compare otherwise) */
+  const char *volatile sym_v = symbol;
+  if (!p && sym_v != NULL && g_self_handle && handle == g_self_handle) {
     tls_no_redirect++;
     if (strncmp(symbol, "rsmi_", 5) == 0) {
       void *h = vgpu_real_rsmi_handle();

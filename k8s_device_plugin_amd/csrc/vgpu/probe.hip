@@ -82,18 +82,18 @@ double vgpu_probe_burn(int iters, int grid, int ms_per_kernel) {
   if (hipMalloc((void **)&sink, sizeof(float)) != hipSuccess) return -1.0;
   uint64_t cycles = (uint64_t)ms_per_kernel * 2400000ULL;
   hipEvent_t t0, t1;
-  hipEventCreate(&t0);
-  hipEventCreate(&t1);
-  hipEventRecord(t0, 0);
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, 0);
   for (int i = 0; i < iters; i++)
     hipLaunchKernelGGL(burn_kernel, dim3(grid), dim3(256), 0, 0, cycles, sink);
-  hipEventRecord(t1, 0);
+  (void)hipEventRecord(t1, 0);
   if (hipEventSynchronize(t1) != hipSuccess) return -1.0;
   float ms = 0;
-  hipEventElapsedTime(&ms, t0, t1);
-  hipEventDestroy(t0);
-  hipEventDestroy(t1);
-  hipFree(sink);
+  (void)hipEventElapsedTime(&ms, t0, t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
+  (void)hipFree(sink);
   return ms / 1000.0;
 }
 
@@ -102,18 +102,18 @@ double vgpu_probe_storm(long n, int grid) {
   float *sink;
   if (hipMalloc((void **)&sink, sizeof(float)) != hipSuccess) return -1.0;
   hipEvent_t t0, t1;
-  hipEventCreate(&t0);
-  hipEventCreate(&t1);
-  hipEventRecord(t0, 0);
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, 0);
   for (long i = 0; i < n; i++)
     hipLaunchKernelGGL(tiny_kernel, dim3(grid), dim3(64), 0, 0, sink);
-  hipEventRecord(t1, 0);
+  (void)hipEventRecord(t1, 0);
   if (hipEventSynchronize(t1) != hipSuccess) return -1.0;
   float ms = 0;
-  hipEventElapsedTime(&ms, t0, t1);
-  hipEventDestroy(t0);
-  hipEventDestroy(t1);
-  hipFree(sink);
+  (void)hipEventElapsedTime(&ms, t0, t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
+  (void)hipFree(sink);
   return ms / 1000.0;
 }
 
