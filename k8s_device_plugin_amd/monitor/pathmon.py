@@ -10,6 +10,7 @@ import glob
 import logging
 import os
 import shutil
+import threading
 import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Set
@@ -36,8 +37,15 @@ class PathMonitor:
         self.containers_dir = os.path.join(hook_path, "containers")
         self.entries: Dict[str, ContainerEntry] = {}
         self.lib_path = lib_path
+        # scan() runs on the monitor loop while live_regions() is called
+        # from gRPC worker threads (noderpc) and the metrics collector
+        self._lock = threading.Lock()
 
     def scan(self, live_pod_uids: Set[str], now: Optional[float] = None) -> None:
+        with self._lock:
+            self._scan_locked(live_pod_uids, now)
+
+    def _scan_locked(self, live_pod_uids: Set[str], now: Optional[float]) -> None:
         now = time.time() if now is None else now
         seen = set()
         for d in glob.glob(os.path.join(self.containers_dir, "*")):
@@ -82,4 +90,5 @@ class PathMonitor:
                     e.region.close()
 
     def live_regions(self) -> List[ContainerEntry]:
-        return [e for e in self.entries.values() if e.region is not None]
+        with self._lock:
+            return [e for e in self.entries.values() if e.region is not None]
