@@ -473,3 +473,99 @@ class TestConcurrentFilter:
         assert len(placed) == 2
         usage, _ = sched.get_nodes_usage(["n1"])
         assert usage["n1"].devices[0].usedmem <= 294912
+
+
+class TestExtenderHTTP:
+    """The actual extender wire: POST /filter /bind /webhook with the JSON
+    shapes kube-scheduler sends (ExtenderArgs/ExtenderBindingArgs,
+    reference routes/route.go:41-134)."""
+
+    def _serve(self):
+        import json as j
+        import urllib.request
+
+        from k8s_device_plugin_amd.device.amd import (
+            HANDSHAKE_ANNO,
+            REGISTER_ANNO,
+        )
+        from k8s_device_plugin_amd.scheduler.core import Scheduler
+        from k8s_device_plugin_amd.scheduler.routes import ExtenderServer
+        from k8s_device_plugin_amd.utils.codec import encode_node_devices
+        from k8s_device_plugin_amd.utils.kubeclient import FakeKubeClient
+        from k8s_device_plugin_amd.utils.types import DeviceInfo, NodeInfo
+
+        client = FakeKubeClient()
+        client.add_node(NodeInfo(name="n1", annotations={
+            HANDSHAKE_ANNO: "Reported 2026-01-01 00:00:00",
+            REGISTER_ANNO: encode_node_devices([DeviceInfo(
+                id="GPU-h", count=10, devmem=294912, devcore=100,
+                type="AMD-Instinct-MI355X", numa=0, health=True, index=0)]),
+        }))
+        sched = Scheduler(client)
+        sched.register_from_node_annotations_once()
+        server = ExtenderServer(sched, host="127.0.0.1", port=0)
+        server.start()
+
+        def post(path, obj):
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{server.port}{path}",
+                data=j.dumps(obj).encode(),
+                headers={"Content-Type": "application/json"})
+            with urllib.request.urlopen(req, timeout=10) as resp:
+                return j.loads(resp.read())
+
+        return client, sched, server, post
+
+    def test_filter_bind_webhook_wire(self):
+        client, sched, server, post = self._serve()
+        try:
+            pod_obj = {
+                "kind": "Pod",
+                "metadata": {"name": "w1", "namespace": "default",
+                             "uid": "uid-w1"},
+                "spec": {"containers": [{
+                    "name": "main",
+                    "resources": {"limits": {"amd.com/gpu": "1",
+                                             "amd.com/gpumem": "1024"}},
+                }]},
+            }
+            from k8s_device_plugin_amd.utils.types import PodInfo
+
+            client.add_pod(PodInfo.from_k8s(pod_obj))
+            out = post("/filter", {"Pod": pod_obj, "NodeNames": ["n1"]})
+            assert out["NodeNames"] == ["n1"]
+            assert not out.get("Error")
+
+            out = post("/bind", {"PodName": "w1", "PodNamespace": "default",
+                                 "Node": "n1"})
+            assert not out.get("Error")
+            assert client.get_pod("w1").annotations.get("amd.io/bind-phase") == "allocating"
+
+            review = {
+                "apiVersion": "admission.k8s.io/v1",
+                "kind": "AdmissionReview",
+                "request": {"uid": "wh1", "object": pod_obj,
+                            "kind": {"kind": "Pod"}},
+            }
+            out = post("/webhook", review)
+            assert out["response"]["allowed"] and out["response"]["uid"] == "wh1"
+        finally:
+            server.stop()
+
+    def test_filter_no_fit_reports_failed_nodes(self):
+        client, sched, server, post = self._serve()
+        try:
+            pod_obj = {
+                "kind": "Pod",
+                "metadata": {"name": "w2", "namespace": "default",
+                             "uid": "uid-w2"},
+                "spec": {"containers": [{
+                    "name": "main",
+                    "resources": {"limits": {"amd.com/gpu": "99"}},
+                }]},
+            }
+            out = post("/filter", {"Pod": pod_obj, "NodeNames": ["n1"]})
+            assert not out.get("NodeNames")
+            assert "n1" in (out.get("FailedNodes") or {})
+        finally:
+            server.stop()
