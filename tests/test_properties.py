@@ -109,3 +109,31 @@ def test_fit_never_overcommits(devstates, nums, memreq, cores):
             assert d.usedmem + p.usedmem <= d.totalmem
             assert d.usedcores + p.usedcores <= d.totalcore
             assert d.used < d.count
+
+
+@given(st.text(alphabet=st.characters(blacklist_categories=("Cs",)),
+               max_size=200))
+@settings(max_examples=200, deadline=None)
+def test_decoders_never_crash_on_garbage(s):
+    """Malformed node/pod annotations must raise CodecError (or parse),
+    never crash with an arbitrary exception — the scheduler ingests these
+    from any node object in the cluster."""
+    from k8s_device_plugin_amd.utils.codec import (
+        CodecError,
+        decode_container_devices,
+        decode_node_devices,
+        decode_pod_devices,
+    )
+    from k8s_device_plugin_amd.utils.types import IN_REQUEST_DEVICES
+
+    for fn in (decode_node_devices, decode_container_devices):
+        try:
+            fn(s)
+        except CodecError:
+            pass
+    try:
+        decode_pod_devices(IN_REQUEST_DEVICES,
+                           {IN_REQUEST_DEVICES["AMD"]: s})
+    except CodecError:
+        pass
+    decode_node_xgmi(s)  # xgmi decode is total: any text -> dict
