@@ -67,3 +67,20 @@ amdsmi_status_t amdsmi_get_gpu_vram_usage(amdsmi_processor_handle h,
   info->vram_used = 200u << 10;
   return 0;
 }
+
+typedef struct {
+  uint32_t gfx_activity;
+  uint32_t umc_activity;
+  uint32_t mm_activity;
+  uint32_t reserved[13];
+} amdsmi_engine_usage_t;
+
+amdsmi_status_t amdsmi_get_gpu_activity(amdsmi_processor_handle h,
+                                        amdsmi_engine_usage_t *info) {
+  (void)h;
+  if (!info) return 1;
+  info->gfx_activity = 90; /* "physically busy" */
+  info->umc_activity = 40;
+  info->mm_activity = 0;
+  return 0;
+}

@@ -24,6 +24,14 @@ extern amdsmi_status_t amdsmi_get_gpu_memory_usage(amdsmi_processor_handle,
                                                    int, uint64_t *);
 extern amdsmi_status_t amdsmi_get_gpu_vram_usage(amdsmi_processor_handle,
                                                  amdsmi_vram_usage_t *);
+typedef struct {
+  uint32_t gfx_activity;
+  uint32_t umc_activity;
+  uint32_t mm_activity;
+  uint32_t reserved[13];
+} amdsmi_engine_usage_t;
+extern amdsmi_status_t amdsmi_get_gpu_activity(amdsmi_processor_handle,
+                                               amdsmi_engine_usage_t *);
 
 int main(void) {
   /* AMDSMI_INIT_AMD_GPUS = 1<<1 (amdsmi.h:51; the fake ignores it) */
@@ -46,10 +54,12 @@ int main(void) {
     amdsmi_get_gpu_memory_total(hs[i], 0, &total);
     amdsmi_get_gpu_memory_usage(hs[i], 0, &used);
     amdsmi_get_gpu_vram_usage(hs[i], &vu);
+    amdsmi_engine_usage_t act = {0};
+    amdsmi_get_gpu_activity(hs[i], &act);
     printf("{\"dev\":%u,\"total\":%llu,\"used\":%llu,"
-           "\"vram_total_mb\":%u,\"vram_used_mb\":%u}\n",
+           "\"vram_total_mb\":%u,\"vram_used_mb\":%u,\"gfx\":%u}\n",
            i, (unsigned long long)total, (unsigned long long)used,
-           vu.vram_total, vu.vram_used);
+           vu.vram_total, vu.vram_used, act.gfx_activity);
   }
   fflush(stdout);
   return 0;

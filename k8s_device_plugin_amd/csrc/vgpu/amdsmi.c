@@ -128,3 +128,28 @@ amdsmi_status_t amdsmi_get_gpu_vram_usage(amdsmi_processor_handle h,
   }
   return s;
 }
+
+typedef struct {
+  uint32_t gfx_activity; /* % */
+  uint32_t umc_activity;
+  uint32_t mm_activity;
+  uint32_t reserved[13];
+} amdsmi_engine_usage_t; /* amdsmi.h:1147-1152 */
+
+amdsmi_status_t amdsmi_get_gpu_activity(amdsmi_processor_handle h,
+                                        amdsmi_engine_usage_t *info) {
+  typedef amdsmi_status_t (*fn)(amdsmi_processor_handle,
+                                amdsmi_engine_usage_t *);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_amdsmi("amdsmi_get_gpu_activity");
+  if (!real) return 1;
+  amdsmi_status_t s = real(h, info);
+  if (s != AMDSMI_OK || vgpu_control_disabled() || !info) return s;
+  int vdev = handle_to_vdev(h);
+  if (vdev >= 0) {
+    uint64_t lim = vgpu_region_sm_limit(vdev);
+    if (lim > 0 && lim < 100 && info->gfx_activity > lim)
+      info->gfx_activity = (uint32_t)lim; /* quota view, as smi.c does */
+  }
+  return s;
+}

@@ -62,3 +62,11 @@ def test_amdsmi_disable_control(tmp_path):
     devs = run_consumer(tmp_path / "r.cache", limit="73728m",
                         extra={"VGPU_DISABLE_CONTROL": "1"})
     assert devs[0]["total"] == 288 * GIB
+
+
+def test_amdsmi_activity_clamped_to_cu_limit(tmp_path):
+    devs = run_consumer(tmp_path / "r.cache", limit="73728m",
+                        extra={"VGPU_DEVICE_CU_LIMIT": "25"})
+    assert devs[0]["gfx"] == 25   # physical 90 clamped to quota
+    devs = run_consumer(tmp_path / "r2.cache", limit="73728m")
+    assert devs[0]["gfx"] == 90   # no CU limit -> passthrough
