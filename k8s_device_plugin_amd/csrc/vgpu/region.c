@@ -173,6 +173,11 @@ static void region_open(void) {
     return;
   }
   vgpu_region_t *r = (vgpu_region_t *)mem;
+  if (!fresh && r->magic == VGPU_MAGIC && r->version != VGPU_VERSION)
+    vgpu_log(VGPU_WARN,
+             "shared region version %u != %u (mixed library versions in one "
+             "container); new fields read as zero-safe defaults",
+             r->version, VGPU_VERSION);
   if (fresh || r->magic != VGPU_MAGIC ||
       __atomic_load_n(&r->init_flag, __ATOMIC_ACQUIRE) != 2) {
     region_init_fields(r);
