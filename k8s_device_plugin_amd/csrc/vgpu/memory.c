@@ -16,6 +16,10 @@
 #define _GNU_SOURCE
 #include "vgpu.h"
 
+#include <fcntl.h>
+#include <sys/file.h>
+#include <unistd.h>
+
 #include <pthread.h>
 #include <stdio.h>
 #include <stdlib.h>
@@ -125,6 +129,31 @@ static int oversubscribe_mode(void) {
   return r && r->oversubscribe;
 }
 
+/* Host-wide serialization of managed (pageable) allocations across
+ * containers (reference unified_lock on /tmp/vgpulock/lock with 1 s wait +
+ * expiry strings, SURVEY.md §2.6 "Oversubscription"): concurrent UVM
+ * carving from several pods thrashes the migration machinery.  flock is
+ * used instead of an expiring lockfile — the kernel releases it when the
+ * holder dies, so no stale-lock repair is needed. */
+static int unified_lock_acquire(void) {
+  int fd = open("/tmp/vgpulock/lock", O_CREAT | O_RDWR, 0666);
+  if (fd < 0) return -1; /* lock dir not mounted: proceed unserialized */
+  for (int tries = 0; tries < 100; tries++) { /* <= 10 s */
+    if (flock(fd, LOCK_EX | LOCK_NB) == 0) return fd;
+    usleep(100 * 1000);
+  }
+  vgpu_log(VGPU_WARN, "unified lock busy >10s; proceeding without it");
+  close(fd);
+  return -1;
+}
+
+static void unified_lock_release(int fd) {
+  if (fd >= 0) {
+    flock(fd, LOCK_UN);
+    close(fd);
+  }
+}
+
 /* ---- hooks ----------------------------------------------------------- */
 typedef hipError_t (*fn_malloc)(void **, size_t);
 typedef hipError_t (*fn_malloc_flags)(void **, size_t, unsigned int);
@@ -147,9 +176,12 @@ static hipError_t alloc_common(void **ptr, size_t size, const char *via,
   hipError_t e;
   vgpu_tls_passthrough++; /* the HIP runtime allocates via HSA underneath */
   if (!vgpu_control_disabled() && oversubscribe_mode()) {
-    /* managed allocation: XNACK pages beyond-HBM working sets to host DRAM */
+    /* managed allocation: XNACK pages beyond-HBM working sets to host DRAM;
+     * serialized host-wide so co-located pods don't thrash UVM */
+    int lk = unified_lock_acquire();
     fn_malloc managed = (fn_malloc)vgpu_real_hip("hipMallocManaged");
     e = managed ? managed(ptr, size) : real_fn(ptr, size);
+    unified_lock_release(lk);
   } else {
     e = real_fn(ptr, size);
   }

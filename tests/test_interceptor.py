@@ -313,3 +313,34 @@ class TestForkSafety:
         assert by_cmd["forkdone"]["status"] == 0
         # after the child exits, liveness pruning returns the memory
         assert by_cmd["meminfo"]["free"] == 1000 * MIB
+
+
+class TestUnifiedLock:
+    def test_managed_alloc_serialized_via_vgpulock(self, tmp_path):
+        """Oversubscribe-mode allocations take the host-wide /tmp/vgpulock
+        flock (reference unified_lock): with a competing holder the alloc
+        waits; without the dir it proceeds unserialized."""
+        import fcntl
+
+        os.makedirs("/tmp/vgpulock", exist_ok=True)
+        lockfile = open("/tmp/vgpulock/lock", "w")
+        fcntl.flock(lockfile, fcntl.LOCK_EX)
+        t0 = time.time()
+        proc = subprocess.Popen(
+            [str(CONSUMER), "allocmanaged", str(64 * MIB)],
+            env={**os.environ,
+                 "LD_LIBRARY_PATH": str(FAKEDIR),
+                 "LD_PRELOAD": str(LIBVGPU),
+                 "VGPU_DEVICE_MEMORY_SHARED_CACHE": str(tmp_path / "r.cache"),
+                 "VGPU_REAL_HIP_PATH": str(FAKEDIR / "libamdhip64.so"),
+                 "VGPU_DEVICE_MEMORY_LIMIT": "1000m",
+                 "VGPU_OVERSUBSCRIBE": "true"},
+            stdout=subprocess.PIPE, text=True)
+        time.sleep(0.6)
+        assert proc.poll() is None, "alloc should be waiting on the lock"
+        fcntl.flock(lockfile, fcntl.LOCK_UN)
+        lockfile.close()
+        out, _ = proc.communicate(timeout=60)
+        assert proc.returncode == 0
+        assert json.loads(out.splitlines()[0])["err"] == 0
+        assert time.time() - t0 >= 0.5
