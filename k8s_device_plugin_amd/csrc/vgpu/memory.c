@@ -221,7 +221,10 @@ hipError_t hipMallocManaged(void **ptr, size_t size, unsigned int flags) {
   if (!vgpu_control_disabled() && vgpu_oom_check(dev, size) != 0)
     return hipErrorOutOfMemory;
   vgpu_tls_passthrough++;
+  int lk = (!vgpu_control_disabled() && oversubscribe_mode())
+               ? unified_lock_acquire() : -1;
   hipError_t e = real_hipMallocManaged(ptr, size, flags);
+  unified_lock_release(lk);
   vgpu_tls_passthrough--;
   if (e == hipSuccess && !vgpu_control_disabled()) account_alloc(*ptr, size, dev);
   return e;

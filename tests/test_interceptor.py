@@ -327,7 +327,7 @@ class TestUnifiedLock:
         fcntl.flock(lockfile, fcntl.LOCK_EX)
         t0 = time.time()
         proc = subprocess.Popen(
-            [str(CONSUMER), "allocmanaged", str(64 * MIB)],
+            [str(CONSUMER), "alloc", str(64 * MIB)],
             env={**os.environ,
                  "LD_LIBRARY_PATH": str(FAKEDIR),
                  "LD_PRELOAD": str(LIBVGPU),
