@@ -94,3 +94,18 @@ def test_scheduler_args_defaults_match_reference():
     assert a.metrics_bind_address == ":9395"
     a = sched_parse_args(["--default-mem", "2048", "--default-cores", "10"])
     assert (a.default_mem, a.default_cores) == (2048, 10)
+
+
+def test_zoo_cli_smoke(capsys, monkeypatch):
+    """The in-cluster benchmark entry point (benchmarks/ai-benchmark Jobs)
+    runs a case end-to-end on CPU."""
+    import torch.nn as nn
+
+    from k8s_device_plugin_amd.models import zoo
+
+    tiny = zoo.BenchCase("tiny", lambda: nn.Linear(8, 4), "inference", 2, (8,))
+    monkeypatch.setitem(zoo.CASES, "tiny", tiny)
+    zoo.main(["--cases", "tiny", "--steps", "3", "--warmup", "1",
+              "--device", "cpu"])
+    out = capsys.readouterr().out
+    assert "tiny" in out and "samples_per_s" in out
