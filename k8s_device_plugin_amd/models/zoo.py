@@ -129,9 +129,13 @@ class ASPP(nn.Module):
                 nn.Conv2d(in_ch, out_ch, 3, padding=r, dilation=r, bias=False),
                 nn.BatchNorm2d(out_ch), nn.ReLU(inplace=True))
                for r in rates])
+        # image-pooling branch: GroupNorm, not BatchNorm — after global
+        # pooling the spatial extent is 1x1, and the published training
+        # config is batch=1 (README.md:253-254), where BatchNorm cannot
+        # compute statistics (1 value/channel) and raises in train mode
         self.gp = nn.Sequential(
             nn.AdaptiveAvgPool2d(1), nn.Conv2d(in_ch, out_ch, 1, bias=False),
-            nn.BatchNorm2d(out_ch), nn.ReLU(inplace=True))
+            nn.GroupNorm(32, out_ch), nn.ReLU(inplace=True))
         self.project = nn.Sequential(
             nn.Conv2d(out_ch * 5, out_ch, 1, bias=False),
             nn.BatchNorm2d(out_ch), nn.ReLU(inplace=True))
