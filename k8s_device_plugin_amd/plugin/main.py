@@ -23,7 +23,7 @@ import threading
 import time
 from typing import Optional
 
-from ..utils.kubeclient import KubeClient, RestKubeClient
+from ..utils.kubeclient import KubeClient, KubeError, RestKubeClient
 from .config import PluginConfig, parse_args
 from .health import HealthChecker
 from .kfd import enumerate_gpus, kfd_healthy
@@ -176,7 +176,11 @@ def main(argv=None) -> int:
         return 2
     if not kfd_healthy():
         log.warning("/dev/kfd not openable — running without enforcement-capable GPUs")
-    daemon = PluginDaemon(cfg)
+    try:
+        daemon = PluginDaemon(cfg)
+    except KubeError as e:
+        log.error("kubernetes API unreachable: %s", e)
+        return 2
     signal.signal(signal.SIGTERM, daemon.request_stop)
     signal.signal(signal.SIGINT, daemon.request_stop)
     signal.signal(signal.SIGHUP, daemon.request_reload)

@@ -17,7 +17,7 @@ import sys
 import threading
 
 from ..device import init_devices
-from ..utils.kubeclient import RestKubeClient
+from ..utils.kubeclient import KubeError, RestKubeClient
 from .core import Scheduler
 from .metrics import metrics_text
 from .routes import ExtenderServer
@@ -60,7 +60,11 @@ def main(argv=None) -> int:
         default_cores=args.default_cores,
     )
     host, _, port = args.http_bind.rpartition(":")
-    client = RestKubeClient()
+    try:
+        client = RestKubeClient()
+    except KubeError as e:
+        log.error("kubernetes API unreachable: %s", e)
+        return 2
     sched = Scheduler(client)
     sched.rebuild_pod_cache()
 
