@@ -18,6 +18,12 @@ test-gpu:
 bench:
 	python bench.py --gpus 1
 
+bench-density:
+	python benchmarks/density_bench.py --pods 10 --prewarm --arbitrate
+
+bench-sched:
+	python benchmarks/sched_bench.py --pods 32 --gpus 8
+
 docker:
 	docker build -f docker/Dockerfile -t $(IMAGE):$(VERSION) .
 
