@@ -92,8 +92,10 @@ hipError_t hipExtMallocWithFlags(void **ptr, size_t size, unsigned flags) {
   return fake_alloc(ptr, size, 0);
 }
 
+static unsigned g_last_managed_flags = 0xdeadbeef;
+
 hipError_t hipMallocManaged(void **ptr, size_t size, unsigned flags) {
-  (void)flags;
+  __atomic_store_n(&g_last_managed_flags, flags, __ATOMIC_RELAXED);
   return fake_alloc(ptr, size, 1);
 }
 
@@ -199,12 +201,128 @@ hipError_t hipModuleLaunchKernel(void *f, unsigned gx, unsigned gy, unsigned gz,
 hipError_t hipDeviceSynchronize(void) { return hipSuccess; }
 hipError_t hipStreamSynchronize(void *s) { (void)s; return hipSuccess; }
 
+/* ---- graph API subset (real HIP semantics: hipGraphGetNodes returns node
+ * handles; hipGraphKernelNodeGetParams fails on non-kernel nodes) -------- */
+typedef struct { /* hipKernelNodeParams ABI (hip_runtime_api.h:1492-1499) */
+  vdim3 blockDim;
+  void **extra;
+  void *func;
+  vdim3 gridDim;
+  void **kernelParams;
+  unsigned int sharedMemBytes;
+} fake_kernel_node_params_t;
+
+#define FAKE_GRAPH_MAX_NODES 20000
+typedef struct {
+  int is_kernel;
+  fake_kernel_node_params_t params;
+} fake_node_t;
+
+typedef struct {
+  fake_node_t *nodes;
+  size_t n;
+} fake_graph_t;
+
+hipError_t hipGraphCreate(void **graph, unsigned int flags) {
+  (void)flags;
+  if (!graph) return hipErrorInvalidValue;
+  fake_graph_t *g = calloc(1, sizeof(fake_graph_t));
+  if (!g) return hipErrorOutOfMemory;
+  g->nodes = calloc(FAKE_GRAPH_MAX_NODES, sizeof(fake_node_t));
+  if (!g->nodes) { free(g); return hipErrorOutOfMemory; }
+  *graph = g;
+  return hipSuccess;
+}
+
+hipError_t hipGraphDestroy(void *graph) {
+  fake_graph_t *g = (fake_graph_t *)graph;
+  if (g) { free(g->nodes); free(g); }
+  return hipSuccess;
+}
+
+hipError_t hipGraphAddKernelNode(void **node, void *graph, const void **deps,
+                                 size_t ndeps,
+                                 const fake_kernel_node_params_t *params) {
+  (void)deps; (void)ndeps;
+  fake_graph_t *g = (fake_graph_t *)graph;
+  if (!g || !params || g->n >= FAKE_GRAPH_MAX_NODES)
+    return hipErrorInvalidValue;
+  g->nodes[g->n].is_kernel = 1;
+  g->nodes[g->n].params = *params;
+  if (node) *node = &g->nodes[g->n];
+  g->n++;
+  return hipSuccess;
+}
+
+hipError_t hipGraphAddEmptyNode(void **node, void *graph, const void **deps,
+                                size_t ndeps) {
+  (void)deps; (void)ndeps;
+  fake_graph_t *g = (fake_graph_t *)graph;
+  if (!g || g->n >= FAKE_GRAPH_MAX_NODES) return hipErrorInvalidValue;
+  g->nodes[g->n].is_kernel = 0;
+  if (node) *node = &g->nodes[g->n];
+  g->n++;
+  return hipSuccess;
+}
+
+hipError_t hipGraphGetNodes(void *graph, void **nodes, size_t *numNodes) {
+  fake_graph_t *g = (fake_graph_t *)graph;
+  if (!g || !numNodes) return hipErrorInvalidValue;
+  if (nodes == NULL) {
+    *numNodes = g->n;
+    return hipSuccess;
+  }
+  size_t n = *numNodes < g->n ? *numNodes : g->n;
+  for (size_t i = 0; i < n; i++) nodes[i] = &g->nodes[i];
+  *numNodes = n;
+  return hipSuccess;
+}
+
+hipError_t hipGraphKernelNodeGetParams(void *node,
+                                       fake_kernel_node_params_t *params) {
+  fake_node_t *fn = (fake_node_t *)node;
+  if (!fn || !params) return hipErrorInvalidValue;
+  if (!fn->is_kernel) return hipErrorInvalidValue;
+  *params = fn->params;
+  return hipSuccess;
+}
+
+hipError_t hipGraphInstantiate(void **pGraphExec, void *graph, void *pErrNode,
+                               char *pLogBuffer, size_t bufferSize) {
+  (void)pErrNode; (void)pLogBuffer; (void)bufferSize;
+  if (!pGraphExec || !graph) return hipErrorInvalidValue;
+  *pGraphExec = graph; /* exec handle aliases the graph in the fake */
+  return hipSuccess;
+}
+
+hipError_t hipGraphInstantiateWithFlags(void **pGraphExec, void *graph,
+                                        unsigned long long flags) {
+  (void)flags;
+  return hipGraphInstantiate(pGraphExec, graph, NULL, NULL, 0);
+}
+
+hipError_t hipGraphLaunch(void *graphExec, void *stream) {
+  (void)stream;
+  fake_graph_t *g = (fake_graph_t *)graphExec;
+  if (!g) return hipErrorInvalidValue;
+  __atomic_fetch_add(&g_launches, g->n, __ATOMIC_RELAXED);
+  return hipSuccess;
+}
+
+hipError_t hipGraphExecDestroy(void *graphExec) {
+  (void)graphExec; /* alias of the graph; freed by hipGraphDestroy */
+  return hipSuccess;
+}
+
 /* test introspection */
 uint64_t fake_hip_launch_count(void) {
   return __atomic_load_n(&g_launches, __ATOMIC_RELAXED);
 }
 uint64_t fake_hip_managed_count(void) {
   return __atomic_load_n(&g_managed_allocs, __ATOMIC_RELAXED);
+}
+uint64_t fake_hip_last_managed_flags(void) {
+  return __atomic_load_n(&g_last_managed_flags, __ATOMIC_RELAXED);
 }
 uint64_t fake_hip_device_usage(int dev) {
   pthread_mutex_lock(&g_mu);

@@ -85,6 +85,23 @@ hsa_status_t hsa_amd_memory_pool_free(void *ptr) {
   return HSA_OK;
 }
 
+/* legacy region API: attrs 0 = SEGMENT (GLOBAL=0 for both),
+ * 1 = GLOBAL_FLAGS (GPU region COARSE_GRAINED=4, CPU FINE_GRAINED=2) —
+ * mirrors hsa.h:3219-3262 so the interceptor's classification is testable */
+hsa_status_t hsa_region_get_info(region_t region, int attr, void *value) {
+  if (!value) return HSA_ERR_INVALID;
+  uint32_t *out = (uint32_t *)value;
+  if (attr == 0) {
+    *out = 0;
+    return HSA_OK;
+  }
+  if (attr == 1) {
+    *out = region.handle == GPU_POOL ? 4u : 2u;
+    return HSA_OK;
+  }
+  return HSA_ERR_INVALID;
+}
+
 hsa_status_t hsa_memory_allocate(region_t region, size_t size, void **ptr) {
   pool_t p = {region.handle};
   return hsa_amd_memory_pool_allocate(p, size, 0, ptr);

@@ -12,6 +12,9 @@
  *   meminfo              hipMemGetInfo
  *   totalmem             hipDeviceTotalMem(dev 0)
  *   launch <n> <grid>    n hipLaunchKernel calls with grid workgroups
+ *   graphlaunch <launches> <nodes> <grid>
+ *                        build a graph of <nodes> kernel nodes (grid wgs
+ *                        each), instantiate, replay it <launches> times
  *   sleep <ms>
  *   setdevice <i>
  */
@@ -35,6 +38,23 @@ extern hipError_t hipDeviceTotalMem(size_t *, int);
 extern hipError_t hipSetDevice(int);
 extern hipError_t hipLaunchKernel(const void *, vdim3, vdim3, void **, size_t,
                                   void *);
+
+typedef struct { /* hipKernelNodeParams ABI */
+  vdim3 blockDim;
+  void **extra;
+  void *func;
+  vdim3 gridDim;
+  void **kernelParams;
+  unsigned int sharedMemBytes;
+} knode_params_t;
+
+extern hipError_t hipGraphCreate(void **, unsigned);
+extern hipError_t hipGraphDestroy(void *);
+extern hipError_t hipGraphAddKernelNode(void **, void *, const void **, size_t,
+                                        const knode_params_t *);
+extern hipError_t hipGraphInstantiate(void **, void *, void *, char *, size_t);
+extern hipError_t hipGraphLaunch(void *, void *);
+extern hipError_t hipGraphExecDestroy(void *);
 
 static double now_s(void) {
   struct timespec ts;
@@ -89,16 +109,47 @@ int main(int argc, char **argv) {
           "{\"cmd\":\"launch\",\"n\":%ld,\"grid\":%u,\"seconds\":%.6f,"
           "\"err\":%d}\n",
           n, grid, dt, e);
+    } else if (strcmp(cmd, "graphlaunch") == 0 && i + 3 < argc) {
+      long launches = atol(argv[++i]);
+      long nodes = atol(argv[++i]);
+      unsigned grid = (unsigned)atoi(argv[++i]);
+      void *graph = NULL, *exec = NULL;
+      hipError_t e = hipGraphCreate(&graph, 0);
+      knode_params_t p;
+      memset(&p, 0, sizeof(p));
+      p.func = (void *)main;
+      p.gridDim.x = grid;
+      p.gridDim.y = p.gridDim.z = 1;
+      p.blockDim.x = 64;
+      p.blockDim.y = p.blockDim.z = 1;
+      for (long k = 0; k < nodes && e == 0; k++)
+        e = hipGraphAddKernelNode(NULL, graph, NULL, 0, &p);
+      if (e == 0) e = hipGraphInstantiate(&exec, graph, NULL, NULL, 0);
+      double t0 = now_s();
+      for (long k = 0; k < launches && e == 0; k++)
+        e = hipGraphLaunch(exec, NULL);
+      double dt = now_s() - t0;
+      if (exec) hipGraphExecDestroy(exec);
+      if (graph) hipGraphDestroy(graph);
+      printf(
+          "{\"cmd\":\"graphlaunch\",\"launches\":%ld,\"nodes\":%ld,"
+          "\"grid\":%u,\"seconds\":%.6f,\"err\":%d}\n",
+          launches, nodes, grid, dt, e);
     } else if (strcmp(cmd, "stats") == 0) {
       /* fake-runtime introspection; -1 on the real runtime */
       long launches = -1, managed = -1;
       typedef unsigned long long (*cnt_fn)(void);
+      long mflags = -1;
       cnt_fn lf = (cnt_fn)dlsym(RTLD_DEFAULT, "fake_hip_launch_count");
       cnt_fn mf = (cnt_fn)dlsym(RTLD_DEFAULT, "fake_hip_managed_count");
+      cnt_fn ff = (cnt_fn)dlsym(RTLD_DEFAULT, "fake_hip_last_managed_flags");
       if (lf) launches = (long)lf();
       if (mf) managed = (long)mf();
-      printf("{\"cmd\":\"stats\",\"launches\":%ld,\"managed\":%ld}\n",
-             launches, managed);
+      if (ff) mflags = (long)ff();
+      printf(
+          "{\"cmd\":\"stats\",\"launches\":%ld,\"managed\":%ld,"
+          "\"managed_flags\":%ld}\n",
+          launches, managed, mflags);
     } else if (strcmp(cmd, "forkhold") == 0 && i + 2 < argc) {
       /* fork(): the child's allocations must account to ITS OWN proc slot
        * (atfork re-registration) and vanish when it exits */

@@ -82,6 +82,14 @@ hipError_t hipMemGetInfo(size_t *free_out, size_t *total_out) {
   }
   uint64_t usage = vgpu_current_usage(dev);
   uint64_t lim_free = usage >= limit ? 0 : limit - usage;
+  /* Co-located pods can leave the card with less physical free than this
+   * container's remaining quota; promising the larger number makes
+   * frameworks over-allocate and hit hard OOM.  Clamp to the real free —
+   * except in oversubscribe mode, where exceeding physical HBM is the
+   * whole point (managed memory pages to host DRAM). */
+  vgpu_region_t *vr = vgpu_region_get();
+  int oversub = vr && vr->oversubscribe;
+  if (!oversub && lim_free > real_free) lim_free = real_free;
   vgpu_log(VGPU_DEBUG,
            "hipMemGetInfo: orig free=%zu total=%zu limit=%llu usage=%llu",
            real_free, real_total, (unsigned long long)limit,

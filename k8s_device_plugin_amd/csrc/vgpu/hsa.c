@@ -164,11 +164,55 @@ hsa_status_t hsa_amd_memory_pool_free(void *ptr) {
   return s;
 }
 
-/* Legacy region-based API: region memory on a GPU agent is device HBM too.
- * Classification by region handle is not portable, so count it only when it
- * succeeds AND the pointer is unknown to the pool ledger; the dominant HSA
- * consumers (HIP runtime, rccl fine-grained staging) use the pool API, and
- * hsa_memory_allocate on ROCm forwards there. */
+/* Legacy region-based API: device HBM shows up here as a GLOBAL-segment
+ * COARSE_GRAINED region (hsa.h:3219-3262 — host staging regions are
+ * FINE_GRAINED/KERNARG), so classify by hsa_region_get_info and charge
+ * coarse-grained global regions to the ledger like the pool path.  The
+ * dominant HSA consumers (HIP runtime, RCCL staging) use the pool API;
+ * this closes the quota bypass for direct legacy-API callers. */
+#define REGION_INFO_SEGMENT 0      /* HSA_REGION_INFO_SEGMENT */
+#define REGION_INFO_GLOBAL_FLAGS 1 /* HSA_REGION_INFO_GLOBAL_FLAGS */
+#define REGION_SEGMENT_GLOBAL 0    /* HSA_REGION_SEGMENT_GLOBAL */
+#define REGION_FLAG_COARSE 4       /* HSA_REGION_GLOBAL_FLAG_COARSE_GRAINED */
+
+static struct { uint64_t handle; int device_local; } g_regions[POOL_CACHE];
+static int g_region_count = 0;
+
+static int legacy_region_is_device_local(hsa_region_t region) {
+  pthread_mutex_lock(&g_pool_mu);
+  for (int i = 0; i < g_region_count; i++) {
+    if (g_regions[i].handle == region.handle) {
+      int r = g_regions[i].device_local;
+      pthread_mutex_unlock(&g_pool_mu);
+      return r;
+    }
+  }
+  pthread_mutex_unlock(&g_pool_mu);
+
+  typedef hsa_status_t (*fn)(hsa_region_t, int, void *);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hsa("hsa_region_get_info");
+  int local = 0;
+  if (real) {
+    uint32_t segment = ~0u, flags = 0;
+    vgpu_tls_passthrough++;
+    hsa_status_t s1 = real(region, REGION_INFO_SEGMENT, &segment);
+    hsa_status_t s2 = real(region, REGION_INFO_GLOBAL_FLAGS, &flags);
+    vgpu_tls_passthrough--;
+    if (s1 == HSA_STATUS_SUCCESS && segment == REGION_SEGMENT_GLOBAL &&
+        s2 == HSA_STATUS_SUCCESS && (flags & REGION_FLAG_COARSE))
+      local = 1;
+  }
+  pthread_mutex_lock(&g_pool_mu);
+  if (g_region_count < POOL_CACHE) {
+    g_regions[g_region_count].handle = region.handle;
+    g_regions[g_region_count].device_local = local;
+    g_region_count++;
+  }
+  pthread_mutex_unlock(&g_pool_mu);
+  return local;
+}
+
 hsa_status_t hsa_memory_allocate(hsa_region_t region, size_t size, void **ptr) {
   typedef hsa_status_t (*fn)(hsa_region_t, size_t, void **);
   static fn real = NULL;
@@ -176,9 +220,23 @@ hsa_status_t hsa_memory_allocate(hsa_region_t region, size_t size, void **ptr) {
   if (!real) return HSA_STATUS_ERROR_OUT_OF_RESOURCES;
   if (vgpu_control_disabled() || vgpu_tls_passthrough)
     return real(region, size, ptr);
+  vgpu_ensure_initialized();
+  int counted = legacy_region_is_device_local(region);
+  int dev = counted ? vgpu_current_device() : -1;
+  if (counted && vgpu_oom_check(dev, size) != 0) {
+    vgpu_log(VGPU_WARN, "hsa region alloc %zu over limit dev=%d", size, dev);
+    return HSA_STATUS_ERROR_OUT_OF_RESOURCES;
+  }
   vgpu_tls_passthrough++;
   hsa_status_t s = real(region, size, ptr);
   vgpu_tls_passthrough--;
+  if (s == HSA_STATUS_SUCCESS && counted) {
+    vgpu_region_t *r = vgpu_region_get();
+    if (r) vgpu_region_add_usage(r, dev, (int64_t)size, 0);
+    hsa_ledger_insert(*ptr, size, dev);
+    vgpu_log(VGPU_DEBUG, "hsa_memory_allocate(%zu) dev=%d -> %p", size, dev,
+             *ptr);
+  }
   return s;
 }
 

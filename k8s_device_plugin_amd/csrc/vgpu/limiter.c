@@ -87,9 +87,10 @@ static void resolve_cards(void) {
                "/sys/class/drm/%s/device/gpu_busy_percent", tok);
     /* gpu_ids still come from KFD below */
   }
-  /* Fallback: map via KFD topology drm_render_minor (card = minor - 128).
-   * Card numbering does NOT follow GPU order on partitioned/multi-VF nodes
-   * (observed on MI355X: one GPU, KFD node 7, render minor 168 = card40). */
+  /* Resolve the DRM card via the device's PCI BDF
+   * (/sys/bus/pci/devices/<bdf>/drm/card*) — card numbering does NOT
+   * follow render_minor-128 on hosts with other DRM devices; the
+   * arithmetic is only the last-ditch fallback. */
   int idx = 0;
   int have_cards = g_card_path[0][0] != 0;
   for (int node = 0; node < 64 && idx < VGPU_MAX_DEVICES; node++) {
@@ -98,12 +99,14 @@ static void resolve_cards(void) {
              "/sys/class/kfd/kfd/topology/nodes/%d/properties", node);
     FILE *f = fopen(p, "r");
     if (!f) continue;
-    long simd = 0, minor = -1;
+    long simd = 0, minor = -1, location = 0, domain = 0;
     char key[64];
     long val;
     while (fscanf(f, "%63s %ld", key, &val) == 2) {
       if (strcmp(key, "simd_count") == 0) simd = val;
       if (strcmp(key, "drm_render_minor") == 0) minor = val;
+      if (strcmp(key, "location_id") == 0) location = val;
+      if (strcmp(key, "domain") == 0) domain = val;
     }
     fclose(f);
     if (simd > 0) {
@@ -117,9 +120,29 @@ static void resolve_cards(void) {
         fclose(gf);
       }
       g_gpu_id[idx] = gid;
-      if (!have_cards && minor >= 128)
-        snprintf(g_card_path[idx], sizeof(g_card_path[0]),
-                 "/sys/class/drm/card%ld/device/gpu_busy_percent", minor - 128);
+      if (!have_cards) {
+        long card = -1;
+        char drm_dir[256];
+        snprintf(drm_dir, sizeof(drm_dir),
+                 "/sys/bus/pci/devices/%04lx:%02lx:%02lx.%lx/drm",
+                 domain, (location >> 8) & 0xff, (location >> 3) & 0x1f,
+                 location & 0x7);
+        DIR *dd = opendir(drm_dir);
+        if (dd) {
+          struct dirent *de;
+          while ((de = readdir(dd)) != NULL)
+            if (strncmp(de->d_name, "card", 4) == 0 && de->d_name[4] >= '0' &&
+                de->d_name[4] <= '9') {
+              card = atol(de->d_name + 4);
+              break;
+            }
+          closedir(dd);
+        }
+        if (card < 0 && minor >= 128) card = minor - 128; /* fallback */
+        if (card >= 0)
+          snprintf(g_card_path[idx], sizeof(g_card_path[0]),
+                   "/sys/class/drm/card%ld/device/gpu_busy_percent", card);
+      }
       idx++;
     }
   }
@@ -264,9 +287,29 @@ static void refill(vgpu_region_t *r, uint64_t now) {
   double dt = (double)(now - last) / NSEC;
   if (dt > 1.0) dt = 1.0;
   double fixed = fixed_rate();
-  /* fdinfo engine time is whole-process (all devices); read it once per
-   * tick, outside the device loop */
-  int self_util = getenv("VGPU_UTIL_FILE") ? -3 : read_self_util_percent();
+  /* fdinfo engine time is whole-process and cannot be attributed to one
+   * device; it is only a valid per-device signal when exactly ONE device
+   * is under a CU limit (the common one-GPU-pod case).  With several
+   * limited devices, feeding the whole-process number to each device's
+   * controller over-throttles all of them — fall back to per-device
+   * cu_occupancy/busy% instead. */
+  int n_limited = 0;
+  for (int d = 0; d < VGPU_MAX_DEVICES; d++)
+    if (r->sm_limit[d] > 0 && r->sm_limit[d] < 100) n_limited++;
+  int self_util;
+  if (getenv("VGPU_UTIL_FILE"))
+    self_util = -3; /* test fixture path */
+  else if (n_limited > 1)
+    self_util = -1; /* per-device signals only */
+  else
+    self_util = read_self_util_percent();
+  /* monitor-scale freshness window: 2.5 x the monitor's own feedback
+   * interval (written into the region), floored at 2 s — the shipped
+   * chart runs the monitor at 5 s, so a fixed 2 s window would expire
+   * between ticks and oscillate against the local EMA controller. */
+  uint64_t mint = __atomic_load_n(&r->monitor_interval_ns, __ATOMIC_RELAXED);
+  uint64_t fresh_win = mint ? mint * 5 / 2 : 2ULL * NSEC;
+  if (fresh_win < 2ULL * NSEC) fresh_win = 2ULL * NSEC;
   for (int d = 0; d < VGPU_MAX_DEVICES; d++) {
     uint64_t lim = r->sm_limit[d];
     if (lim == 0 || lim >= 100) continue;
@@ -276,7 +319,7 @@ static void refill(vgpu_region_t *r, uint64_t now) {
      * multiplier on its entitled share (proportional fairness) */
     uint64_t mts = __atomic_load_n(&r->monitor_scale_ts_ns, __ATOMIC_RELAXED);
     int monitor_fresh = fixed <= 0 && mts != 0 && now > mts &&
-                        now - mts < 2ULL * NSEC;
+                        now - mts < fresh_win;
     if (monitor_fresh) {
       int64_t fp = __atomic_load_n(&r->monitor_scale_fp[d], __ATOMIC_RELAXED);
       if (fp > 0) base *= (double)fp / 1e6;
@@ -375,13 +418,27 @@ void vgpu_limiter_gate(int dev, uint64_t workgroups) {
 
   /* priority gate: the monitor writes recent_kernel = -1 to suspend
    * lower-priority containers while a high-priority one is active
-   * (reference feedback.go:197-255) */
+   * (reference feedback.go:197-255).  The wait is bounded (default 60 s,
+   * VGPU_PRIORITY_WAIT_MS to override) so a crashed monitor can never
+   * deadlock the container; expiry is logged because it means a
+   * high-priority pod's exclusivity lapsed. */
+  static int wait_budget_ms = -1;
+  if (wait_budget_ms < 0) {
+    const char *w = getenv("VGPU_PRIORITY_WAIT_MS");
+    wait_budget_ms = w ? atoi(w) : 60000;
+    if (wait_budget_ms < 0) wait_budget_ms = 0;
+  }
   int waited_ms = 0;
   while (__atomic_load_n(&r->recent_kernel, __ATOMIC_RELAXED) < 0 &&
-         waited_ms < 60000) {
+         waited_ms < wait_budget_ms) {
     usleep(1000);
     waited_ms++;
   }
+  if (waited_ms >= wait_budget_ms && wait_budget_ms > 0 &&
+      __atomic_load_n(&r->recent_kernel, __ATOMIC_RELAXED) < 0)
+    vgpu_log(VGPU_WARN,
+             "priority gate expired after %d ms; proceeding despite block "
+             "(monitor stale or high-priority task overran)", waited_ms);
   int32_t rk = __atomic_load_n(&r->recent_kernel, __ATOMIC_RELAXED);
   if (rk < 100) __atomic_store_n(&r->recent_kernel, rk + 1, __ATOMIC_RELAXED);
 
@@ -394,7 +451,11 @@ void vgpu_limiter_gate(int dev, uint64_t workgroups) {
 
   int64_t cost = (int64_t)workgroups;
   if (cost < 1) cost = 1;
-  if (cost > 65536) cost = 65536; /* one giant grid shouldn't starve forever */
+  /* Overdraw semantics make big costs safe (the bucket goes negative and
+   * later launches pay it off), so the clamp only guards absurd values —
+   * it must be well above a real graph replay's workgroup count or
+   * capture-heavy workloads (torch.compile + hipGraphs) get undercharged. */
+  if (cost > (1 << 20)) cost = 1 << 20;
   for (;;) {
     int64_t cur = __atomic_load_n(&r->core_tokens[dev], __ATOMIC_RELAXED);
     if (cur > 0) {
@@ -408,6 +469,141 @@ void vgpu_limiter_gate(int dev, uint64_t workgroups) {
 
 /* ---- launch hooks ---------------------------------------------------- */
 typedef struct { unsigned x, y, z; } vdim3;
+
+/* ---- graph workgroup accounting ---------------------------------------
+ * A replayed graph bundles many kernel launches and the per-node hooks do
+ * not fire on replay, so hipGraphLaunch must charge the graph's REAL
+ * workgroup count — a flat constant would let a capture-heavy workload
+ * (torch.compile + hipGraphs) blow through its CU limit.  The count is
+ * computed once at hipGraphInstantiate/hipGraphExecUpdate time by walking
+ * the graph's kernel nodes (hipGraphKernelNodeGetParams gridDim), then
+ * looked up per launch. */
+typedef struct {           /* ABI prefix of hipKernelNodeParams
+                            * (hip_runtime_api.h:1492-1499) */
+  vdim3 blockDim;
+  void **extra;
+  void *func;
+  vdim3 gridDim;
+  void **kernelParams;
+  unsigned int sharedMemBytes;
+} vgpu_kernel_node_params_t;
+
+#define GRAPH_COST_SLOTS 512
+#define GRAPH_MAX_NODES 65536
+static struct { void *exec; uint64_t cost; } g_graph_cost[GRAPH_COST_SLOTS];
+static pthread_mutex_t g_graph_mu = PTHREAD_MUTEX_INITIALIZER;
+
+static uint64_t graph_workgroups(void *graph) {
+  typedef hipError_t (*fn_nodes)(void *, void **, size_t *);
+  typedef hipError_t (*fn_kparams)(void *, vgpu_kernel_node_params_t *);
+  static fn_nodes get_nodes = NULL;
+  static fn_kparams get_kparams = NULL;
+  if (!get_nodes) get_nodes = (fn_nodes)vgpu_real_hip("hipGraphGetNodes");
+  if (!get_kparams)
+    get_kparams = (fn_kparams)vgpu_real_hip("hipGraphKernelNodeGetParams");
+  if (!get_nodes || !get_kparams || !graph) return 0;
+  size_t n = 0;
+  if (get_nodes(graph, NULL, &n) != hipSuccess || n == 0) return 0;
+  if (n > GRAPH_MAX_NODES) n = GRAPH_MAX_NODES;
+  void **nodes = malloc(n * sizeof(void *));
+  if (!nodes) return 0;
+  uint64_t cost = 0;
+  if (get_nodes(graph, nodes, &n) == hipSuccess) {
+    for (size_t i = 0; i < n; i++) {
+      vgpu_kernel_node_params_t p;
+      memset(&p, 0, sizeof(p));
+      /* non-kernel nodes (memcpy/memset/empty) return an error: skip */
+      if (get_kparams(nodes[i], &p) == hipSuccess) {
+        uint64_t g = (uint64_t)(p.gridDim.x ? p.gridDim.x : 1) *
+                     (p.gridDim.y ? p.gridDim.y : 1) *
+                     (p.gridDim.z ? p.gridDim.z : 1);
+        cost += g;
+      }
+    }
+  }
+  free(nodes);
+  return cost;
+}
+
+static void graph_cost_set(void *exec, uint64_t cost) {
+  if (!exec) return;
+  pthread_mutex_lock(&g_graph_mu);
+  int free_slot = -1;
+  for (int i = 0; i < GRAPH_COST_SLOTS; i++) {
+    if (g_graph_cost[i].exec == exec) { free_slot = i; break; }
+    if (free_slot < 0 && g_graph_cost[i].exec == NULL) free_slot = i;
+  }
+  if (free_slot >= 0) {
+    g_graph_cost[free_slot].exec = exec;
+    g_graph_cost[free_slot].cost = cost;
+  }
+  pthread_mutex_unlock(&g_graph_mu);
+}
+
+static void graph_cost_drop(void *exec) {
+  pthread_mutex_lock(&g_graph_mu);
+  for (int i = 0; i < GRAPH_COST_SLOTS; i++)
+    if (g_graph_cost[i].exec == exec) {
+      g_graph_cost[i].exec = NULL;
+      g_graph_cost[i].cost = 0;
+      break;
+    }
+  pthread_mutex_unlock(&g_graph_mu);
+}
+
+static uint64_t graph_cost_get(void *exec) {
+  uint64_t c = 0;
+  pthread_mutex_lock(&g_graph_mu);
+  for (int i = 0; i < GRAPH_COST_SLOTS; i++)
+    if (g_graph_cost[i].exec == exec) { c = g_graph_cost[i].cost; break; }
+  pthread_mutex_unlock(&g_graph_mu);
+  return c;
+}
+
+hipError_t hipGraphInstantiate(void **pGraphExec, void *graph, void *pErrorNode,
+                               char *pLogBuffer, size_t bufferSize) {
+  typedef hipError_t (*fn)(void **, void *, void *, char *, size_t);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipGraphInstantiate");
+  if (!real) return hipErrorInvalidValue;
+  hipError_t e = real(pGraphExec, graph, pErrorNode, pLogBuffer, bufferSize);
+  if (e == hipSuccess && pGraphExec)
+    graph_cost_set(*pGraphExec, graph_workgroups(graph));
+  return e;
+}
+
+hipError_t hipGraphInstantiateWithFlags(void **pGraphExec, void *graph,
+                                        unsigned long long flags) {
+  typedef hipError_t (*fn)(void **, void *, unsigned long long);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipGraphInstantiateWithFlags");
+  if (!real) return hipErrorInvalidValue;
+  hipError_t e = real(pGraphExec, graph, flags);
+  if (e == hipSuccess && pGraphExec)
+    graph_cost_set(*pGraphExec, graph_workgroups(graph));
+  return e;
+}
+
+hipError_t hipGraphExecUpdate(void *hGraphExec, void *hGraph,
+                              void **hErrorNode_out, void *updateResult_out) {
+  typedef hipError_t (*fn)(void *, void *, void **, void *);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipGraphExecUpdate");
+  if (!real) return hipErrorInvalidValue;
+  hipError_t e = real(hGraphExec, hGraph, hErrorNode_out, updateResult_out);
+  if (e == hipSuccess) graph_cost_set(hGraphExec, graph_workgroups(hGraph));
+  return e;
+}
+
+hipError_t hipGraphExecDestroy(void *graphExec) {
+  typedef hipError_t (*fn)(void *);
+  static fn real = NULL;
+  if (!real) real = (fn)vgpu_real_hip("hipGraphExecDestroy");
+  if (!real) return hipErrorInvalidValue;
+  hipError_t e = real(graphExec);
+  if (e == hipSuccess) graph_cost_drop(graphExec);
+  return e;
+}
 
 hipError_t hipLaunchKernel(const void *f, vdim3 grid, vdim3 block, void **args,
                            size_t shared, void *stream) {
@@ -482,10 +678,12 @@ hipError_t hipGraphLaunch(void *graphExec, void *stream) {
   if (!real) real = (fn)vgpu_real_hip("hipGraphLaunch");
   if (!real) return hipErrorInvalidValue;
   vgpu_ensure_initialized();
-  /* a replayed graph bundles many launches; charge a graph-sized constant
-   * (the per-node hooks do not fire on replay) */
-  if (!vgpu_control_disabled())
-    vgpu_limiter_gate(vgpu_current_device(), 2048);
+  if (!vgpu_control_disabled()) {
+    /* charge the graph's measured workgroup count (recorded at
+     * instantiate/update time); unknown execs fall back to a constant */
+    uint64_t cost = graph_cost_get(graphExec);
+    vgpu_limiter_gate(vgpu_current_device(), cost ? cost : 2048);
+  }
   return real(graphExec, stream);
 }
 

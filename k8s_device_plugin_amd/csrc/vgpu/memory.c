@@ -177,10 +177,14 @@ static hipError_t alloc_common(void **ptr, size_t size, const char *via,
   vgpu_tls_passthrough++; /* the HIP runtime allocates via HSA underneath */
   if (!vgpu_control_disabled() && oversubscribe_mode()) {
     /* managed allocation: XNACK pages beyond-HBM working sets to host DRAM;
-     * serialized host-wide so co-located pods don't thrash UVM */
+     * serialized host-wide so co-located pods don't thrash UVM.
+     * hipMallocManaged takes THREE arguments — calling it through a 2-arg
+     * pointer leaves flags as register garbage (UB). */
     int lk = unified_lock_acquire();
-    fn_malloc managed = (fn_malloc)vgpu_real_hip("hipMallocManaged");
-    e = managed ? managed(ptr, size) : real_fn(ptr, size);
+    fn_malloc_flags managed =
+        (fn_malloc_flags)vgpu_real_hip("hipMallocManaged");
+    e = managed ? managed(ptr, size, 0x01 /* hipMemAttachGlobal */)
+                : real_fn(ptr, size);
     unified_lock_release(lk);
   } else {
     e = real_fn(ptr, size);

@@ -157,6 +157,27 @@ static void region_open(void) {
 
   struct stat st;
   fstat(fd, &st);
+  /* Version safety BEFORE any ftruncate: a file that already carries our
+   * magic but a different version (mixed library versions in a container,
+   * or an old region file surviving a library upgrade) must not be
+   * re-initialized — that would wipe live processes' accounting — and must
+   * not be read through mismatched offsets.  Refuse loudly; enforcement
+   * for THIS process is then off (fail-open, matching hook-less behavior),
+   * which is observable in the logs rather than silently corrupting. */
+  if (st.st_size >= (off_t)(2 * sizeof(uint32_t))) {
+    uint32_t head[2] = {0, 0};
+    if (pread(fd, head, sizeof(head), 0) == (ssize_t)sizeof(head) &&
+        head[0] == VGPU_MAGIC && head[1] != VGPU_VERSION) {
+      vgpu_log(VGPU_ERR,
+               "shared region %s has version %u but this library is v%u; "
+               "refusing to attach (enforcement disabled for this process; "
+               "redeploy with matching library versions)",
+               path, head[1], VGPU_VERSION);
+      flock(fd, LOCK_UN);
+      close(fd);
+      return;
+    }
+  }
   int fresh = st.st_size < (off_t)sizeof(vgpu_region_t);
   if (fresh && ftruncate(fd, sizeof(vgpu_region_t)) != 0) {
     vgpu_log(VGPU_ERR, "ftruncate %s failed: %s", path, strerror(errno));
@@ -173,11 +194,6 @@ static void region_open(void) {
     return;
   }
   vgpu_region_t *r = (vgpu_region_t *)mem;
-  if (!fresh && r->magic == VGPU_MAGIC && r->version != VGPU_VERSION)
-    vgpu_log(VGPU_WARN,
-             "shared region version %u != %u (mixed library versions in one "
-             "container); new fields read as zero-safe defaults",
-             r->version, VGPU_VERSION);
   if (fresh || r->magic != VGPU_MAGIC ||
       __atomic_load_n(&r->init_flag, __ATOMIC_ACQUIRE) != 2) {
     region_init_fields(r);
@@ -294,7 +310,8 @@ int vgpu_region_layout_json(char *buf, size_t buflen) {
       "\"owner_pid\":%lu,\"num_devices\":%lu,\"uuids\":%lu,\"limit\":%lu,"
       "\"sm_limit\":%lu,\"core_tokens\":%lu,\"token_fill_rate\":%lu,"
       "\"last_refill_ns\":%lu,\"monitor_scale_fp\":%lu,"
-      "\"monitor_scale_ts_ns\":%lu,\"procs\":%lu,\"proc_num\":%lu,"
+      "\"monitor_scale_ts_ns\":%lu,\"monitor_interval_ns\":%lu,"
+      "\"procs\":%lu,\"proc_num\":%lu,"
       "\"utilization_switch\":%lu,\"recent_kernel\":%lu,\"priority\":%lu,"
       "\"oversubscribe\":%lu,\"_proc_slot_size\":%lu,\"_proc_pid\":%lu,"
       "\"_proc_host_pid\":%lu,\"_proc_used\":%lu,\"_proc_monitor_used\":%lu,"
@@ -304,7 +321,7 @@ int vgpu_region_layout_json(char *buf, size_t buflen) {
       OFF(init_flag), OFF(owner_pid), OFF(num_devices), OFF(uuids),
       OFF(limit), OFF(sm_limit), OFF(core_tokens), OFF(token_fill_rate),
       OFF(last_refill_ns), OFF(monitor_scale_fp), OFF(monitor_scale_ts_ns),
-      OFF(procs), OFF(proc_num),
+      OFF(monitor_interval_ns), OFF(procs), OFF(proc_num),
       OFF(utilization_switch), OFF(recent_kernel), OFF(priority),
       OFF(oversubscribe), (unsigned long)sizeof(vgpu_proc_slot_t),
       (unsigned long)offsetof(vgpu_proc_slot_t, pid),

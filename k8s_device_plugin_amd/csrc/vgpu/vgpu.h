@@ -27,7 +27,12 @@ extern "C" {
 #endif
 
 #define VGPU_MAGIC 0x4D495655u /* "MIVU" */
-#define VGPU_VERSION 2
+/* v3: monitor_scale_* moved AFTER procs[] (appending new fields at the end
+ * keeps every pre-existing offset stable across versions) and
+ * monitor_interval_ns added.  A version mismatch on attach now REFUSES the
+ * region (fail-open, loud) instead of re-initializing it — a v2 library
+ * attaching a v3 file must never wipe live accounting. */
+#define VGPU_VERSION 3
 #define VGPU_MAX_DEVICES 16
 #define VGPU_MAX_PROCS 1024
 #define VGPU_UUID_LEN 96
@@ -87,19 +92,22 @@ typedef struct {
   int64_t core_tokens[VGPU_MAX_DEVICES];
   int64_t token_fill_rate[VGPU_MAX_DEVICES]; /* tokens/sec, feedback-adjusted */
   uint64_t last_refill_ns;
-  /* Monitor-arbitrated throttle scale (fixed-point x1e6): the node monitor
-   * sees EVERY container's region, so it writes one per-device scale to all
-   * of them — equal multiplier x entitled CU share = proportional fairness
-   * without per-process utilization attribution (which the kernel cannot
-   * provide for KFD queues).  The limiter uses it while fresh (<2 s). */
-  int64_t monitor_scale_fp[VGPU_MAX_DEVICES];
-  uint64_t monitor_scale_ts_ns;
   vgpu_proc_slot_t procs[VGPU_MAX_PROCS];
   int32_t proc_num;
   int32_t utilization_switch; /* monitor: 1 = enforce CU limit, 0 = free-run */
   int32_t recent_kernel;      /* monitor feedback: <0 = blocked (priority) */
   int32_t priority;           /* this container's task priority */
   uint64_t oversubscribe;     /* 1 = managed-memory alloc mode */
+  /* ---- v3 additions (appended; earlier offsets unchanged) ------------- */
+  /* Monitor-arbitrated throttle scale (fixed-point x1e6): the node monitor
+   * sees EVERY container's region, so it writes one per-device scale to all
+   * of them — equal multiplier x entitled CU share = proportional fairness
+   * without per-process utilization attribution (which the kernel cannot
+   * provide for KFD queues).  The limiter honors it while fresh: younger
+   * than 2.5 x monitor_interval_ns (>= 2 s floor). */
+  int64_t monitor_scale_fp[VGPU_MAX_DEVICES];
+  uint64_t monitor_scale_ts_ns;
+  uint64_t monitor_interval_ns; /* written by the monitor; 0 = unknown */
 } vgpu_region_t;
 
 /* ---- region API (region.c) ---- */

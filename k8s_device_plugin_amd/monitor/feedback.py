@@ -32,15 +32,19 @@ BUSY_HIGH, BUSY_LOW = 95, 85
 
 class FeedbackLoop:
     def __init__(self, pathmon: PathMonitor, soft_cores: bool = False,
-                 busy_reader: Optional[Callable[[str], int]] = None):
+                 busy_reader: Optional[Callable[[str], int]] = None,
+                 interval_s: float = 5.0):
         """soft_cores=True: enforce the CU limit only while the device is
         contended (the reference's default GPU_CORE_UTILIZATION_POLICY);
         False (our default): strict isolation, limit always enforced.
         busy_reader(uuid) -> device busy percent enables node-arbitrated
-        fair throttling (monitor writes one scale to every region)."""
+        fair throttling (monitor writes one scale to every region).
+        interval_s is this loop's period; it is published into every
+        region so the limiter sizes its scale-freshness window to it."""
         self.pathmon = pathmon
         self.soft_cores = soft_cores
         self.busy_reader = busy_reader
+        self.interval_s = interval_s
         self._scale: Dict[str, float] = {}  # device uuid -> multiplier
 
     def observe_once(self) -> None:
@@ -82,6 +86,10 @@ class FeedbackLoop:
                         blocked.add(e.key)
 
         for e in entries:
+            try:
+                e.region.set_monitor_interval(self.interval_s)
+            except Exception:
+                pass  # v2 region without the field
             if e.key in blocked:
                 e.region.set_recent_kernel(-1)
             elif e.region.get_recent_kernel() < 0:

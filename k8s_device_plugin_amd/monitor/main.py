@@ -88,7 +88,8 @@ def main(argv=None):
             return -1
 
     feedback = FeedbackLoop(pathmon, soft_cores=args.soft_cores,
-                            busy_reader=busy_reader if gpus else None)
+                            busy_reader=busy_reader if gpus else None,
+                            interval_s=args.interval)
     collector = MonitorCollector(pathmon, gpus)
     serve_metrics(collector, args.metrics_port)
     if args.grpc_bind:

@@ -115,12 +115,25 @@ class SharedRegion:
 
     def set_monitor_scale(self, dev: int, scale: float, now_ns: int) -> None:
         """Write the node-arbitrated throttle scale (fixed-point x1e6) and
-        freshness timestamp; the limiter uses it while < 2 s old."""
+        freshness timestamp; the limiter honors it while younger than
+        2.5x the monitor interval (set_monitor_interval)."""
         struct.pack_into("<q", self._mm,
                          self.layout["monitor_scale_fp"] + 8 * dev,
                          int(scale * 1e6))
         struct.pack_into("<Q", self._mm,
                          self.layout["monitor_scale_ts_ns"], now_ns)
+
+    def set_monitor_interval(self, interval_s: float) -> None:
+        """Publish the monitor's feedback period so the limiter can size
+        its freshness window (2.5x interval); without this a 5 s monitor
+        would expire a 2 s window for 3 of every 5 seconds and the limiter
+        would oscillate between arbitrated and local control."""
+        struct.pack_into("<Q", self._mm, self.layout["monitor_interval_ns"],
+                         int(interval_s * 1e9))
+
+    def get_monitor_interval(self) -> float:
+        return struct.unpack_from(
+            "<Q", self._mm, self.layout["monitor_interval_ns"])[0] / 1e9
 
     def get_monitor_scale(self, dev: int) -> float:
         v = struct.unpack_from(

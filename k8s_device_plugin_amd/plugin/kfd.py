@@ -48,9 +48,16 @@ class PhysicalGPU:
     # XCD is its own KFD node; amdgpu exposes the mode per drm card)
     compute_partition: str = "SPX"   # SPX / DPX / QPX / CPX
     memory_partition: str = "NPS1"   # NPS1 / NPS2 / NPS4
+    # DRM card index resolved via the device's PCI BDF symlink
+    # (/sys/bus/pci/devices/<bdf>/drm/card*); -1 if unresolvable.  Card
+    # numbering is NOT guaranteed to be render_minor-128 on hosts with
+    # other DRM devices — the arithmetic is only a fallback.
+    drm_card_no: int = -1
 
     @property
     def drm_card(self) -> int:
+        if self.drm_card_no >= 0:
+            return self.drm_card_no
         return self.drm_render_minor - 128
 
     @property
@@ -92,6 +99,17 @@ def _read_str(path: str, default: str = "") -> str:
             return f.read().strip()
     except OSError:
         return default
+
+
+def _resolve_drm_card(pci_root: str, bdf: str) -> int:
+    """DRM card index for a PCI device, from its sysfs drm/ directory —
+    authoritative, unlike render_minor-128 arithmetic (other DRM devices
+    on the host shift card numbering)."""
+    for card in glob.glob(os.path.join(pci_root, bdf, "drm", "card*")):
+        name = os.path.basename(card)
+        if name.startswith("card") and name[4:].isdigit():
+            return int(name[4:])
+    return -1
 
 
 def enumerate_gpus(
@@ -140,7 +158,10 @@ def enumerate_gpus(
             if "node_to" in lprops:
                 io_links[lprops["node_to"]] = lprops.get("type", 0)
         render_minor = props.get("drm_render_minor", 128 + len(gpus))
-        card_dev = os.path.join(drm_root, f"card{render_minor - 128}", "device")
+        card_no = _resolve_drm_card(pci_root, bdf)
+        if card_no < 0:
+            card_no = render_minor - 128  # fallback arithmetic
+        card_dev = os.path.join(drm_root, f"card{card_no}", "device")
         compute_part = _read_str(
             os.path.join(card_dev, "current_compute_partition"), "SPX") or "SPX"
         memory_part = _read_str(
@@ -160,6 +181,7 @@ def enumerate_gpus(
                 io_links=io_links,
                 compute_partition=compute_part,
                 memory_partition=memory_part,
+                drm_card_no=_resolve_drm_card(pci_root, bdf),
             )
         )
     return gpus

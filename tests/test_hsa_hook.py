@@ -95,6 +95,29 @@ class TestHSAQuota:
         assert lines[:2] == ["ok 0", "ok 1"]
 
 
+class TestLegacyRegionAPI:
+    """hsa_memory_allocate on a GLOBAL+COARSE_GRAINED region is device HBM
+    and must hit the same quota ledger (the passthrough left a
+    bypass); host fine-grained regions stay uncounted."""
+
+    def test_legacy_gpu_region_counted(self, tmp_path):
+        lines = run_consumer(["legacy", 600, 600], tmp_path / "r.cache",
+                             mem_limit="1000m")
+        assert lines[0] == "ok 0"
+        assert lines[1] == "oom 1"
+
+    def test_legacy_cpu_region_uncounted(self, tmp_path):
+        lines = run_consumer(["legacycpu", 600, 600], tmp_path / "r.cache",
+                             mem_limit="1000m")
+        assert lines[:2] == ["ok 0", "ok 1"]
+
+    def test_legacy_disable_control_passthrough(self, tmp_path):
+        lines = run_consumer(["legacy", 600, 600], tmp_path / "r.cache",
+                             mem_limit="1000m",
+                             extra_env={"VGPU_DISABLE_CONTROL": "1"})
+        assert lines[:2] == ["ok 0", "ok 1"]
+
+
 def test_exports_present():
     """The .so exports the HSA hook symbols (so PLT interposition works)."""
     out = subprocess.run(["nm", "-D", str(LIBVGPU)], capture_output=True,
