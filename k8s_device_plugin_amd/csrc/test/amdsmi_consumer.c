@@ -1,7 +1,24 @@
 /* PLT-linked amd-smi-style consumer: enumerates processors and prints the
- * memory view per device — run under LD_PRELOAD to verify quota spoofing. */
+ * memory view per device — run under LD_PRELOAD to verify quota spoofing.
+ *
+ * AMDSMI_BT=1 installs a SIGBUS/SIGSEGV handler that prints a backtrace
+ * before dying (triage tooling for the real-library preload crash,
+ * ROUND2_NOTES item 5). */
+#define _GNU_SOURCE
+#include <execinfo.h>
+#include <signal.h>
 #include <stdint.h>
 #include <stdio.h>
+#include <stdlib.h>
+#include <unistd.h>
+
+static void crash_handler(int sig) {
+  void *frames[64];
+  int n = backtrace(frames, 64);
+  dprintf(2, "FATAL signal %d; backtrace (%d frames):\n", sig, n);
+  backtrace_symbols_fd(frames, n, 2);
+  _exit(128 + sig);
+}
 
 typedef int amdsmi_status_t;
 typedef void *amdsmi_processor_handle;
@@ -34,6 +51,10 @@ extern amdsmi_status_t amdsmi_get_gpu_activity(amdsmi_processor_handle,
                                                amdsmi_engine_usage_t *);
 
 int main(void) {
+  if (getenv("AMDSMI_BT")) {
+    signal(SIGBUS, crash_handler);
+    signal(SIGSEGV, crash_handler);
+  }
   /* AMDSMI_INIT_AMD_GPUS = 1<<1 (amdsmi.h:51; the fake ignores it) */
   amdsmi_init(1 << 1);
   amdsmi_socket_handle sockets[8];

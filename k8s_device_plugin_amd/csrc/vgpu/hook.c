@@ -125,6 +125,7 @@ void vgpu_ensure_initialized(void) { pthread_once(&g_init_once, do_init); }
 int vgpu_initialized(void) { return !vgpu_control_disabled(); }
 
 /* ---- dlopen interposition ------------------------------------------- */
+#ifndef VGPU_NO_DLOPEN
 static const char *self_path(void) {
   static char path[4096];
   if (!path[0]) {
@@ -163,10 +164,12 @@ void *dlopen(const char *filename, int flags) {
   }
   return real_dlopen()(filename, flags);
 }
+#endif /* VGPU_NO_DLOPEN */
 
 /* dlsym interposition: a redirected handle must still resolve the hundreds
  * of symbols we do not hook — forward misses to the real library, chosen by
  * symbol prefix. */
+#ifndef VGPU_NO_DLSYM
 typedef void *(*dlsym_fn)(void *, const char *);
 static dlsym_fn real_dlsym(void) {
   static dlsym_fn fn = NULL;
@@ -205,3 +208,4 @@ void *dlsym(void *handle, const char *symbol) {
   }
   return p;
 }
+#endif /* VGPU_NO_DLSYM */
