@@ -8,23 +8,30 @@ libvgpu-hip.so with a 50% HBM quota, the published job shape
 (/root/reference/benchmarks/ai-benchmark/Hami/ai-benchmark.yml:14-19:
 gpumem-percentage: 50).  Reported value = overhead percent (lower better).
 
-Per rank (one GPU, launched by torch.distributed.run for N>1):
-  1. probe the GPU's HBM size (subprocess, no preload);
-  2. native phase: worker subprocess builds the cases, warms up W steps,
-     signals READY; parent barriers all ranks (gloo), sends GO; worker times
-     EXACTLY K steps with torch.cuda.synchronize around the timed region;
-  3. vgpu phase: identical worker, plus LD_PRELOAD + VGPU_DEVICE_MEMORY_LIMIT
-     = quota% of HBM + a fresh shared region;
-  4. all-reduce: aggregate throughputs (sum) and step time (max) over ranks.
+Phases per rank (one GPU, launched by torch.distributed.run for N>1):
+  1. native     — one bare worker, whole GPU;
+  2. vgpu       — one worker under LD_PRELOAD + 50% HBM quota (the direct
+                  enforcement overhead, the headline value);
+  3. colocated  — TWO workers under LD_PRELOAD at 50% quota each, run
+                  CONCURRENTLY on the same GPU (the published 2-pods/GPU
+                  job shape); aggregate vs native reported in config;
+  4. density    — N=1 only: 10 resnet50_inf pods at 10% CU / 10% HBM each
+                  with monitor-style arbitration, reporting aggregate and
+                  fairness spread (reference deviceSplitCount=10 claim).
 
 A "step" = one iteration of every selected ai-benchmark case (default:
-ResNet-V2-50 inference b50@346^2 + training b20@346^2; --cases all runs the
-full 10-case suite).  Synthetic data, random-init weights, fp32 (the
-reference suite's dtype).
+the full 10-case suite, README.md:243-256).  Synthetic data, random-init
+weights, fp32 (the reference suite's dtype).
+
+MIOpen: if miopen_udb/ (pre-tuned find-db, captured on MI355X) exists in
+the repo it is copied to a writable tmp dir and exported, so runs skip
+the multi-minute cold autotune and use tuned kernels (not naive_conv
+fallbacks) in both phases.
 """
 import argparse
 import json
 import os
+import shutil
 import subprocess
 import sys
 import tempfile
@@ -33,6 +40,8 @@ from pathlib import Path
 
 REPO = Path(__file__).resolve().parent
 LIBVGPU = REPO / "k8s_device_plugin_amd" / "csrc" / "libvgpu-hip.so"
+MIOPEN_UDB = REPO / "miopen_udb"
+MIOPEN_CACHE = REPO / "miopen_cache"
 
 
 def parse_args():
@@ -40,10 +49,16 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--cases", default="resnet50_inf,resnet50_train",
-                   help="comma list or 'all'")
+    p.add_argument("--cases", default="all",
+                   help="comma list or 'all' (the 10-case suite)")
     p.add_argument("--quota-pct", type=int, default=50)
+    p.add_argument("--skip-colocated", action="store_true")
+    p.add_argument("--density-pods", type=int, default=10,
+                   help="pods in the density phase (0 disables; N=1 only)")
+    p.add_argument("--density-seconds", type=float, default=20.0)
     p.add_argument("--worker", action="store_true", help=argparse.SUPPRESS)
+    p.add_argument("--density-worker", action="store_true",
+                   help=argparse.SUPPRESS)
     p.add_argument("--device", type=int, default=0, help=argparse.SUPPRESS)
     return p.parse_args()
 
@@ -56,10 +71,50 @@ def resolve_cases(spec):
     return [c.strip() for c in spec.split(",") if c.strip()]
 
 
+def miopen_env(tmp_root):
+    """Writable copies of the shipped pre-tuned MIOpen dbs (if present)."""
+    env = {}
+    if MIOPEN_UDB.is_dir():
+        dst = os.path.join(tmp_root, "miopen_udb")
+        if not os.path.isdir(dst):
+            shutil.copytree(MIOPEN_UDB, dst)
+        env["MIOPEN_USER_DB_PATH"] = dst
+    if MIOPEN_CACHE.is_dir():
+        dst = os.path.join(tmp_root, "miopen_cache")
+        if not os.path.isdir(dst):
+            shutil.copytree(MIOPEN_CACHE, dst)
+        env["MIOPEN_CUSTOM_CACHE_DIR"] = dst
+    return env
+
+
 # ---------------------------------------------------------------------------
-# Worker: runs inside the (optionally preloaded) subprocess.
+# Workers: run inside the (optionally preloaded) subprocess.
 # ---------------------------------------------------------------------------
+FAKE_GPU = os.environ.get("BENCH_FAKE_GPU") == "1"
+"""CPU test mode: exercises the FULL multi-rank orchestration (rendezvous,
+READY/GO choreography, phases, aggregation, JSON contract) with fake
+workers — so the driver's first 8-GPU SCALE run is not the first execution
+of this code path.  Never used for reported numbers."""
+
+
 def worker_main(args):
+    if FAKE_GPU:
+        from k8s_device_plugin_amd.models import zoo
+
+        names = resolve_cases(args.cases)
+        samples_per_step = sum(zoo.CASES[c].batch for c in names)
+        print("READY", flush=True)
+        assert sys.stdin.readline().strip() == "GO"
+        time.sleep(0.05 * args.steps)
+        elapsed = 0.05 * args.steps
+        print(json.dumps({
+            "elapsed_s": elapsed,
+            "steps": args.steps,
+            "samples_per_step": samples_per_step,
+            "samples_per_s": samples_per_step * args.steps / elapsed,
+        }), flush=True)
+        return
+
     import torch
 
     from k8s_device_plugin_amd.models import zoo
@@ -103,10 +158,44 @@ def worker_main(args):
     }), flush=True)
 
 
+def density_worker_main(args):
+    if FAKE_GPU:
+        print("READY", flush=True)
+        assert sys.stdin.readline().strip() == "GO"
+        time.sleep(min(args.density_seconds, 0.2))
+        print(json.dumps({"samples_per_s": 100.0, "steps": 1}), flush=True)
+        return
+
+    import torch
+
+    from k8s_device_plugin_amd.models import zoo
+
+    case = zoo.CASES["resnet50_inf"]
+    dev = torch.device("cuda", 0)
+    model = zoo.build(case, dev)
+    batch = zoo.synthetic_batch(case, dev)
+    for _ in range(3):
+        zoo.step(case, model, batch, None)
+    torch.cuda.synchronize()
+    print("READY", flush=True)
+    assert sys.stdin.readline().strip() == "GO"
+    t0 = time.perf_counter()
+    steps = 0
+    while time.perf_counter() - t0 < args.density_seconds:
+        zoo.step(case, model, batch, None)
+        steps += 1
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(json.dumps({"samples_per_s": case.batch * steps / dt,
+                      "steps": steps}), flush=True)
+
+
 # ---------------------------------------------------------------------------
 # Parent: per-rank orchestration.
 # ---------------------------------------------------------------------------
 def probe_total_mem(device_env):
+    if FAKE_GPU:
+        return 288 << 30
     code = ("import torch;"
             "print(torch.cuda.get_device_properties(0).total_memory)")
     out = subprocess.run([sys.executable, "-c", code], env=device_env,
@@ -116,35 +205,156 @@ def probe_total_mem(device_env):
     return int(out.stdout.strip().splitlines()[-1])
 
 
-def run_phase(args, device_env, barrier):
+def run_phase(args, envs, barrier):
+    """Launch len(envs) workers concurrently; barrier across ranks when ALL
+    local workers are warm; return the list of result dicts."""
     cmd = [sys.executable, str(REPO / "bench.py"), "--worker",
            "--cases", args.cases, "--steps", str(args.steps),
            "--warmup", str(args.warmup)]
-    proc = subprocess.Popen(cmd, env=device_env, stdin=subprocess.PIPE,
-                            stdout=subprocess.PIPE, text=True, cwd=str(REPO))
-    result = None
+    procs = [subprocess.Popen(cmd, env=e, stdin=subprocess.PIPE,
+                              stdout=subprocess.PIPE, text=True,
+                              cwd=str(REPO)) for e in envs]
+    results = [None] * len(procs)
     try:
-        for line in proc.stdout:
-            if line.strip() == "READY":
-                barrier()  # all ranks' workers are warm
-                proc.stdin.write("GO\n")
-                proc.stdin.flush()
-            elif line.startswith("{"):
-                result = json.loads(line)
-        proc.wait(timeout=60)
+        for i, proc in enumerate(procs):
+            line = proc.stdout.readline()
+            assert line.strip() == "READY", \
+                f"worker {i} failed before READY: {line!r}"
+        barrier()  # all ranks' workers are warm
+        for proc in procs:
+            proc.stdin.write("GO\n")
+            proc.stdin.flush()
+        for i, proc in enumerate(procs):
+            for line in proc.stdout:
+                if line.startswith("{"):
+                    results[i] = json.loads(line)
+                    break
+            proc.wait(timeout=120)
     finally:
-        if proc.poll() is None:
-            proc.kill()
+        for proc in procs:
+            if proc.poll() is None:
+                proc.kill()
     barrier()  # all ranks finished the timed region
-    if result is None:
-        raise RuntimeError("worker produced no result")
-    return result
+    for i, r in enumerate(results):
+        if r is None:
+            raise RuntimeError(f"worker {i} produced no result")
+    return results
+
+
+def vgpu_env(base_env, tmp_root, quota_mib, tag, cu_limit=None):
+    cache = tempfile.NamedTemporaryFile(
+        prefix=f"vgpu-bench-{tag}-", suffix=".cache", delete=False,
+        dir=tmp_root)
+    cache.close()
+    env = dict(base_env)
+    env.update({
+        "LD_PRELOAD": str(LIBVGPU),
+        "VGPU_DEVICE_MEMORY_LIMIT": f"{quota_mib}m",
+        "VGPU_DEVICE_MEMORY_SHARED_CACHE": cache.name,
+        "VGPU_DEVICE_UUIDS": f"GPU-bench-{tag}",
+    })
+    if cu_limit:
+        env["VGPU_DEVICE_CU_LIMIT"] = str(cu_limit)
+    return env
+
+
+def run_density(args, base_env, tmp_root, total_mem):
+    """10 co-located inference pods at 10% HBM / 10% CU with the monitor's
+    arbitration loop live (the production fair-sharing stack)."""
+    import threading
+
+    from k8s_device_plugin_amd.monitor.region import SharedRegion
+
+    pods = args.density_pods
+    quota_mib = total_mem // pods // (1 << 20)
+    cu_pct = max(1, 100 // pods)
+    envs = [vgpu_env(base_env, tmp_root, quota_mib, f"density-{i}",
+                     cu_limit=cu_pct) for i in range(pods)]
+    caches = [e["VGPU_DEVICE_MEMORY_SHARED_CACHE"] for e in envs]
+    cmd = [sys.executable, str(REPO / "bench.py"), "--density-worker",
+           "--density-seconds", str(args.density_seconds)]
+    procs = [subprocess.Popen(cmd, env=e, stdin=subprocess.PIPE,
+                              stdout=subprocess.PIPE, text=True,
+                              cwd=str(REPO)) for e in envs]
+
+    # monitor-style arbitration: one AIMD scale for the device, written to
+    # every pod's region (monitor/feedback.py _arbitrate, in-process here
+    # because there is no monitor daemon on a bench box)
+    stop = threading.Event()
+
+    def busy_path():
+        import glob as g
+        cards = sorted(g.glob("/sys/class/drm/card*/device/gpu_busy_percent"))
+        return cards[0] if cards else None
+
+    def arbitrate():
+        regions = {}
+        scale = 1.0
+        path = busy_path()
+        while not stop.is_set():
+            for c in caches:
+                if c not in regions:
+                    try:
+                        r = SharedRegion(c)
+                        if r.valid:
+                            regions[c] = r
+                    except (OSError, ValueError):
+                        pass
+            busy = -1
+            if path:
+                try:
+                    busy = int(open(path).read().strip())
+                except (OSError, ValueError):
+                    pass
+            if busy > 95:
+                scale = max(0.05, scale * 0.90)
+            elif 0 <= busy < 85:
+                scale = min(100.0, scale * 1.10)
+            now = time.monotonic_ns()
+            for r in regions.values():
+                try:
+                    r.set_monitor_interval(0.25)
+                    r.set_monitor_scale(0, scale, now)
+                except (OSError, ValueError):
+                    pass
+            stop.wait(0.25)
+
+    threading.Thread(target=arbitrate, daemon=True).start()
+    try:
+        for proc in procs:
+            assert proc.stdout.readline().strip() == "READY"
+        for proc in procs:
+            proc.stdin.write("GO\n")
+            proc.stdin.flush()
+        results = []
+        for proc in procs:
+            line = ""
+            for line in proc.stdout:
+                if line.startswith("{"):
+                    break
+            proc.wait(timeout=180)
+            results.append(json.loads(line))
+    finally:
+        stop.set()
+        for proc in procs:
+            if proc.poll() is None:
+                proc.kill()
+    rates = [r["samples_per_s"] for r in results]
+    return {
+        "pods": pods,
+        "aggregate_samples_per_s": round(sum(rates), 2),
+        "per_pod_samples_per_s": [round(r, 2) for r in rates],
+        "fairness_max_over_min": round(max(rates) / max(min(rates), 1e-9), 3),
+    }
 
 
 def main():
     args = parse_args()
     if args.worker:
         worker_main(args)
+        return
+    if args.density_worker:
+        density_worker_main(args)
         return
 
     rank = int(os.environ.get("RANK", "0"))
@@ -162,37 +372,51 @@ def main():
         if dist is not None:
             dist.barrier()
 
+    tmp_root = tempfile.mkdtemp(prefix=f"vgpu-bench-r{rank}-")
     base_env = dict(os.environ)
     # one GPU per rank; the worker sees it as cuda:0
     base_env["CUDA_VISIBLE_DEVICES"] = str(local_rank)
     base_env["HIP_VISIBLE_DEVICES"] = str(local_rank)
     base_env.pop("LD_PRELOAD", None)
+    base_env.update(miopen_env(tmp_root))
     # keep workers out of the parent's rendezvous
     for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT",
               "GROUP_RANK", "LOCAL_WORLD_SIZE", "TORCHELASTIC_RUN_ID"):
         base_env.pop(k, None)
+
+    if not LIBVGPU.exists():
+        raise RuntimeError(f"{LIBVGPU} missing — run __graft_entry__.build()")
 
     total_mem = probe_total_mem(base_env)
     quota_bytes = total_mem * args.quota_pct // 100
     quota_mib = quota_bytes // (1 << 20)
 
     t_job0 = time.perf_counter()
-    native = run_phase(args, base_env, barrier)
+    native = run_phase(args, [base_env], barrier)[0]
 
-    vgpu_env = dict(base_env)
-    cache = tempfile.NamedTemporaryFile(prefix="vgpu-bench-", suffix=".cache",
-                                        delete=False)
-    cache.close()
-    vgpu_env.update({
-        "LD_PRELOAD": str(LIBVGPU),
-        "VGPU_DEVICE_MEMORY_LIMIT": f"{quota_mib}m",
-        "VGPU_DEVICE_MEMORY_SHARED_CACHE": cache.name,
-        "VGPU_DEVICE_UUIDS": f"GPU-bench-{local_rank}",
-    })
-    if not LIBVGPU.exists():
-        raise RuntimeError(f"{LIBVGPU} missing — run __graft_entry__.build()")
-    vgpu = run_phase(args, vgpu_env, barrier)
-    os.unlink(cache.name)
+    vgpu = run_phase(
+        args, [vgpu_env(base_env, tmp_root, quota_mib, f"r{rank}")],
+        barrier)[0]
+
+    colocated = None
+    if not args.skip_colocated:
+        colo = run_phase(
+            args,
+            [vgpu_env(base_env, tmp_root, quota_mib, f"r{rank}c{i}")
+             for i in range(2)],
+            barrier)
+        colocated = {
+            "workers": 2,
+            "aggregate_samples_per_s":
+                round(sum(r["samples_per_s"] for r in colo), 2),
+            "per_worker_samples_per_s":
+                [round(r["samples_per_s"], 2) for r in colo],
+        }
+
+    density = None
+    if args.density_pods > 0 and world == 1:
+        density = run_density(args, base_env, tmp_root, total_mem)
+
     wall = time.perf_counter() - t_job0
 
     import torch
@@ -213,7 +437,36 @@ def main():
     vgpu_ms_per_step = agg(vgpu["elapsed_s"] * 1000.0 / args.steps, MAX)
     overhead_pct = (native_total - vgpu_total) / native_total * 100.0
 
+    colo_total = None
+    colo_overhead = None
+    if colocated is not None:
+        colo_total = agg(colocated["aggregate_samples_per_s"], SUM)
+        colo_overhead = (native_total - colo_total) / native_total * 100.0
+
+    shutil.rmtree(tmp_root, ignore_errors=True)
+
     if rank == 0:
+        config = {
+            "model": "ai-benchmark:" + args.cases,
+            "global_batch": sum(
+                __import__("k8s_device_plugin_amd.models.zoo",
+                           fromlist=["CASES"]).CASES[c].batch
+                for c in resolve_cases(args.cases)) * world,
+            "seq_len": None,
+            "parallelism": f"pods-per-gpu quota={args.quota_pct}%mem",
+            "quota_mib": quota_mib,
+            "native_samples_per_s": round(native_total, 2),
+            "vgpu_samples_per_s": round(vgpu_total, 2),
+            "enforcement": "LD_PRELOAD libvgpu-hip.so, hard HBM cap",
+            "miopen_db": "shipped" if MIOPEN_UDB.is_dir() else "cold",
+            "wall_s": round(wall, 1),
+        }
+        if colocated is not None:
+            config["colocated_2pods"] = colocated
+            config["colocated_aggregate_samples_per_s"] = round(colo_total, 2)
+            config["colocated_overhead_pct"] = round(colo_overhead, 3)
+        if density is not None:
+            config["density"] = density
         print(json.dumps({
             "metric": "vGPU overhead vs native (%) on ai-benchmark",
             "value": round(overhead_pct, 3),
@@ -227,19 +480,7 @@ def main():
             "vs_baseline": None,
             "dtype": "fp32",
             "data": "synthetic",
-            "config": {
-                "model": "ai-benchmark:" + args.cases,
-                "global_batch": sum(
-                    __import__("k8s_device_plugin_amd.models.zoo",
-                               fromlist=["CASES"]).CASES[c].batch
-                    for c in resolve_cases(args.cases)) * world,
-                "seq_len": None,
-                "parallelism": f"pods-per-gpu quota={args.quota_pct}%mem",
-                "quota_mib": quota_mib,
-                "native_samples_per_s": round(native_total, 2),
-                "vgpu_samples_per_s": round(vgpu_total, 2),
-                "enforcement": "LD_PRELOAD libvgpu-hip.so, hard HBM cap",
-            },
+            "config": config,
         }), flush=True)
     if dist is not None:
         dist.destroy_process_group()
