@@ -42,6 +42,10 @@ class PluginConfig:
     # reference --device-list-strategy, main.go:61-70)
     device_list_strategy: str = "envvar"
     cdi_spec_dir: str = "/var/run/cdi"
+    # desired compute-partition mode applied at startup: "keep" (default),
+    # or SPX/DPX/QPX/CPX — CPX advertises each XCD as a 32-CU hard-isolated
+    # device (the MIG-strategy analog; plugin/partition.py)
+    compute_partition: str = "keep"
 
 
 def parse_args(argv: Optional[List[str]] = None) -> PluginConfig:
@@ -61,6 +65,8 @@ def parse_args(argv: Optional[List[str]] = None) -> PluginConfig:
                    choices=["envvar", "cdi-annotations"])
     p.add_argument("--context-overhead-mb", type=int, default=c.context_overhead_mb)
     p.add_argument("--cdi-spec-dir", default=c.cdi_spec_dir)
+    p.add_argument("--compute-partition", default=c.compute_partition,
+                   choices=["keep", "SPX", "DPX", "QPX", "CPX"])
     a = p.parse_args(argv)
     cfg = PluginConfig(
         node_name=a.node_name,
@@ -76,6 +82,7 @@ def parse_args(argv: Optional[List[str]] = None) -> PluginConfig:
         device_list_strategy=a.device_list_strategy,
         cdi_spec_dir=a.cdi_spec_dir,
         context_overhead_mb=a.context_overhead_mb,
+        compute_partition=a.compute_partition,
     )
     return apply_node_config(cfg)
 
@@ -99,5 +106,6 @@ def apply_node_config(cfg: PluginConfig) -> PluginConfig:
                 device_split_count=int(entry.get("devicesplitcount", cfg.device_split_count)),
                 device_memory_scaling=float(entry.get("devicememoryscaling", cfg.device_memory_scaling)),
                 device_cores_scaling=float(entry.get("devicecorescaling", cfg.device_cores_scaling)),
+                compute_partition=str(entry.get("computepartition", cfg.compute_partition)),
             )
     return cfg

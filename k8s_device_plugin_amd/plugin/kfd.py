@@ -53,6 +53,11 @@ class PhysicalGPU:
     # numbering is NOT guaranteed to be render_minor-128 on hosts with
     # other DRM devices — the arithmetic is only a fallback.
     drm_card_no: int = -1
+    # CPX bookkeeping: partitions of one physical card share unique_id;
+    # partition_index disambiguates (0..7), parent_uuid groups them
+    partition_index: int = 0
+    partition_count: int = 1     # siblings sharing the physical card
+    parent_uuid: str = ""        # == uuid when unpartitioned
 
     @property
     def drm_card(self) -> int:
@@ -184,7 +189,33 @@ def enumerate_gpus(
                 drm_card_no=_resolve_drm_card(pci_root, bdf),
             )
         )
+    _disambiguate_partitions(gpus)
     return gpus
+
+
+def _disambiguate_partitions(gpus: List[PhysicalGPU]) -> None:
+    """CPX: the 8 XCD partitions of one card share amdgpu's unique_id, so
+    the raw UUIDs collide.  Suffix each sibling with its partition index
+    (GPU-<id>.<k>) — the MIG-instance-UUID analog — and, when siblings all
+    report the SAME memory banks (CPX+NPS1: one shared HBM view per
+    partition), divide the advertised capacity so the card's memory is not
+    counted 8x."""
+    by_uid: Dict[str, List[PhysicalGPU]] = {}
+    for g in gpus:
+        by_uid.setdefault(g.uuid, []).append(g)
+    for uid, group in by_uid.items():
+        if len(group) == 1:
+            g = group[0]
+            g.parent_uuid = g.uuid
+            continue
+        shared_view = len({g.mem_bytes for g in group}) == 1
+        for k, g in enumerate(sorted(group, key=lambda g: g.node_id)):
+            g.parent_uuid = uid
+            g.partition_index = k
+            g.partition_count = len(group)
+            g.uuid = f"{uid}.{k}"
+            if shared_view and g.mem_bytes:
+                g.mem_bytes //= len(group)
 
 
 def kfd_healthy(kfd_dev: str = KFD_DEV) -> bool:

@@ -35,6 +35,14 @@ log = logging.getLogger(__name__)
 
 
 def build_resource_manager(cfg: PluginConfig) -> ResourceManager:
+    if cfg.compute_partition.upper() != "KEEP":
+        from . import partition
+
+        if not partition.apply_mode(cfg.compute_partition):
+            log.error(
+                "could not apply compute partition %s (GPU busy or "
+                "unsupported); advertising current partitioning",
+                cfg.compute_partition)
     gpus = enumerate_gpus()
     if not gpus:
         log.warning("no AMD GPUs found in KFD topology")

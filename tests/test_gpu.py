@@ -47,8 +47,15 @@ def preload_env(tmp_path, limit=None, extra=None):
 
 class TestProbeKernels:
     def test_vecadd_numerics(self):
-        probe = ctypes.CDLL(str(LIBPROBE))
-        assert probe.vgpu_probe_vecadd(ctypes.c_size_t(1 << 22)) == 0
+        # subprocess, not in-process CDLL: the pytest process must never
+        # initialize HIP, or the CPX partition switch below gets EBUSY
+        code = (
+            "import ctypes, json;"
+            f"p = ctypes.CDLL('{LIBPROBE}');"
+            "rc = p.vgpu_probe_vecadd(ctypes.c_size_t(1 << 22));"
+            "print(json.dumps({'rc': rc}))"
+        )
+        assert run_child(code, {})["rc"] == 0
 
     def test_vecadd_numerics_under_preload(self, tmp_path):
         code = (
@@ -175,6 +182,70 @@ class TestCUMaskReal:
         if ratio < 1.5:
             pytest.skip(f"HSA_CU_MASK unsupported on this ASIC (ratio {ratio:.2f})")
         assert ratio > 2.0
+
+
+class TestCPXPartitionReal:
+    def test_cpx_hard_isolation(self):
+        """The hard CU-isolation path: switch the card
+        to CPX (8 XCD-GPUs), run two GPU burns either on DISJOINT
+        partitions or crammed onto the SAME partition, and require the
+        shared-partition co-run to be measurably slower — i.e. disjoint
+        partitions isolate.  Restores the original mode afterwards."""
+        import concurrent.futures as cf
+
+        from k8s_device_plugin_amd.plugin import partition as pt
+        from k8s_device_plugin_amd.plugin.kfd import enumerate_gpus
+
+        files = pt.partition_files()
+        if not files:
+            pytest.skip("no compute-partition sysfs on this host")
+        orig = pt.read_mode(files[0])
+        if orig is None:
+            pytest.skip("compute-partition mode unreadable")
+        if orig != "CPX" and not pt.write_mode(files[0], "CPX"):
+            pytest.skip("cannot switch to CPX (busy or unsupported)")
+        try:
+            gpus = enumerate_gpus()
+            parts = [g for g in gpus if g.partition_count > 1]
+            if len(parts) < 2:
+                # some stacks keep unique_id distinct per partition; fall
+                # back on counting 32-CU nodes
+                parts = [g for g in gpus if g.cu_count <= 64]
+            assert len(parts) >= 2, \
+                f"CPX mode but {len(parts)} partitions enumerated"
+
+            burn = ("import ctypes, json, sys;"
+                    f"p = ctypes.CDLL('{LIBPROBE}');"
+                    "p.vgpu_probe_burn.restype = ctypes.c_double;"
+                    "t = p.vgpu_probe_burn(12, 64, 20);"
+                    "print(json.dumps({'t': t}))")
+
+            def run_burn(visible):
+                return run_child(burn, {"ROCR_VISIBLE_DEVICES": visible},
+                                 timeout=300)["t"]
+
+            def co_run(vis_a, vis_b):
+                with cf.ThreadPoolExecutor(2) as ex:
+                    fa = ex.submit(run_burn, vis_a)
+                    fb = ex.submit(run_burn, vis_b)
+                    return max(fa.result(), fb.result())
+
+            solo = run_burn("0")
+            disjoint = co_run("0", "1")       # two pods, two XCDs
+            shared = co_run("0", "0")         # two pods, one XCD
+            assert solo > 0 and disjoint > 0 and shared > 0
+            # disjoint partitions: hardware isolation -> each run is close
+            # to solo speed; same partition: the two burns serialize
+            assert shared > 1.5 * disjoint, (
+                f"no isolation: solo={solo:.2f} disjoint={disjoint:.2f} "
+                f"shared={shared:.2f}")
+            assert disjoint < 1.4 * solo, (
+                f"disjoint partitions interfere: solo={solo:.2f} "
+                f"disjoint={disjoint:.2f}")
+        finally:
+            if orig and orig != "CPX":
+                assert pt.write_mode(files[0], orig), \
+                    f"FAILED to restore partition mode {orig}"
 
 
 class TestLimiterReal:
