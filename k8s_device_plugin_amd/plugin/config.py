@@ -46,6 +46,11 @@ class PluginConfig:
     # or SPX/DPX/QPX/CPX — CPX advertises each XCD as a 32-CU hard-isolated
     # device (the MIG-strategy analog; plugin/partition.py)
     compute_partition: str = "keep"
+    # time-slicing/replica config (reference rm/device_map.go:37-317):
+    # per-device replica counts overriding device_split_count, plus an
+    # optional resource rename.  Populated from the node JSON's
+    # "timeslicing" block; device key "*" matches all.
+    replica_overrides: dict = field(default_factory=dict)
 
 
 def parse_args(argv: Optional[List[str]] = None) -> PluginConfig:
@@ -101,11 +106,44 @@ def apply_node_config(cfg: PluginConfig) -> PluginConfig:
     for entry in data.get("nodeconfig", []):
         if entry.get("name") == cfg.node_name:
             log.info("applying node config override for %s", cfg.node_name)
-            return replace(
+            cfg = replace(
                 cfg,
                 device_split_count=int(entry.get("devicesplitcount", cfg.device_split_count)),
                 device_memory_scaling=float(entry.get("devicememoryscaling", cfg.device_memory_scaling)),
                 device_cores_scaling=float(entry.get("devicecorescaling", cfg.device_cores_scaling)),
                 compute_partition=str(entry.get("computepartition", cfg.compute_partition)),
             )
-    return cfg
+            break
+    return apply_time_slicing(cfg, data.get("timeslicing") or {})
+
+
+def apply_time_slicing(cfg: PluginConfig, ts: dict) -> PluginConfig:
+    """Reference rm/device_map.go:37-317 replica machinery, MI355X-sized:
+
+    "timeslicing": {"resources": [
+        {"name": "amd.com/gpu", "rename": "amd.com/gpu.shared",
+         "replicas": 20, "devices": ["GPU-abc", ...]}   # devices optional
+    ]}
+
+    A matching entry overrides the fan-out (replicas) for the listed
+    devices ("*" / omitted = all) and may rename the advertised resource.
+    """
+    overrides = dict(cfg.replica_overrides)
+    resource = cfg.resource_name
+    for entry in ts.get("resources", []):
+        if entry.get("name") and entry["name"] != cfg.resource_name:
+            continue
+        replicas = int(entry.get("replicas", 0))
+        if replicas <= 0:
+            log.warning("timeslicing entry without positive replicas: %r",
+                        entry)
+            continue
+        for dev in entry.get("devices") or ["*"]:
+            overrides[str(dev)] = replicas
+        if entry.get("rename"):
+            resource = str(entry["rename"])
+            log.info("timeslicing renames resource %s -> %s",
+                     cfg.resource_name, resource)
+    if overrides == cfg.replica_overrides and resource == cfg.resource_name:
+        return cfg
+    return replace(cfg, replica_overrides=overrides, resource_name=resource)

@@ -57,6 +57,7 @@ def build_resource_manager(cfg: PluginConfig) -> ResourceManager:
         split_count=cfg.device_split_count,
         memory_scaling=cfg.device_memory_scaling,
         cores_scaling=cfg.device_cores_scaling,
+        replica_overrides=cfg.replica_overrides,
     )
 
 

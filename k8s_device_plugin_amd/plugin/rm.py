@@ -30,13 +30,22 @@ class FakeDevice:
 class ResourceManager:
     def __init__(self, gpus: List[PhysicalGPU], split_count: int = 10,
                  memory_scaling: float = 1.0, cores_scaling: float = 1.0,
-                 device_type: str = MI355X_DEVICE_TYPE):
+                 device_type: str = MI355X_DEVICE_TYPE,
+                 replica_overrides: Optional[Dict[str, int]] = None):
         self.gpus = gpus
         self.split_count = max(1, split_count)
         self.memory_scaling = memory_scaling
         self.cores_scaling = cores_scaling
         self.device_type = device_type
+        # per-device replica counts from the time-slicing config block
+        # (reference rm/device_map.go:37-317); "*" matches every device
+        self.replica_overrides = dict(replica_overrides or {})
         self.health: Dict[str, bool] = {g.uuid: True for g in gpus}
+
+    def replicas_for(self, uuid: str) -> int:
+        ov = self.replica_overrides
+        n = ov.get(uuid) or ov.get("*") or self.split_count
+        return max(1, int(n))
 
     def by_uuid(self, uuid: str) -> Optional[PhysicalGPU]:
         for g in self.gpus:
@@ -47,7 +56,7 @@ class ResourceManager:
     def fake_devices(self) -> List[FakeDevice]:
         out: List[FakeDevice] = []
         for g in self.gpus:
-            for i in range(self.split_count):
+            for i in range(self.replicas_for(g.uuid)):
                 out.append(FakeDevice(
                     id=f"{g.uuid}-{i}",
                     uuid=g.uuid,
@@ -79,7 +88,7 @@ class ResourceManager:
                 else f"{self.device_type}-{part}"
             out.append(DeviceInfo(
                 id=g.uuid,
-                count=self.split_count,
+                count=self.replicas_for(g.uuid),
                 devmem=int(g.mem_bytes / MIB * self.memory_scaling),
                 devcore=int(100 * self.cores_scaling),
                 type=dtype,
