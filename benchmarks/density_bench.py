@@ -183,6 +183,7 @@ def main():
                 now = time.monotonic_ns()
                 for r in regions.values():
                     try:
+                        r.set_monitor_interval(0.25)
                         r.set_monitor_scale(0, scale, now)
                     except (OSError, ValueError):
                         pass
