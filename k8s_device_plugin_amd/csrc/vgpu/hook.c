@@ -78,11 +78,36 @@ void *vgpu_real_rsmi_handle(void) {
   return g_real_rsmi;
 }
 
+/* rsmi symbol resolution, ODR-safe order (amd-smi crash triage,
+ * gpurun_out/amdsmi_triage_r2.log): libamd_smi EMBEDS the amd::smi C++
+ * classes that librocm_smi64 also defines, with layouts that differ
+ * between the two builds — dlopening librocm_smi64 into a process that
+ * already has libamd_smi loaded makes librocm_smi's static initializers
+ * bind to libamd_smi's incompatible copies and SIGBUS.  So: resolve
+ * through RTLD_NEXT first (the implementation already in the process —
+ * libamd_smi exports the whole rsmi_* surface), and only dlopen the real
+ * librocm_smi64 when the process has none (the rocm-smi CLI dlopen path,
+ * where no libamd_smi is resident and the load is safe). */
+void *vgpu_real_rsmi_sym(const char *sym) {
+  if (!getenv(ENV_REAL_RSMI)) { /* test override must stay authoritative */
+    void *p = dlsym(RTLD_NEXT, sym);
+    if (p) return p;
+  }
+  void *h = vgpu_real_rsmi_handle();
+  return h ? dlsym(h, sym) : NULL;
+}
+
 __thread int vgpu_tls_passthrough = 0;
 
 static void *g_real_amdsmi = NULL;
 
 void *vgpu_real_amdsmi(const char *sym) {
+  /* same ODR hazard in reverse (librocm_smi64 resident, then dlopen
+   * libamd_smi): prefer the copy already linked into the process */
+  if (!getenv(ENV_REAL_AMDSMI)) {
+    void *p = dlsym(RTLD_NEXT, sym);
+    if (p) return p;
+  }
   if (!g_real_amdsmi) {
     static const char *const names[] = {"libamd_smi.so.26", "libamd_smi.so.25",
                                         "libamd_smi.so", NULL};
@@ -148,15 +173,11 @@ void *dlopen(const char *filename, int flags) {
         void *h = real_dlopen()(self, flags);
         if (h) {
           g_self_handle = h;
-          /* make sure the real library is resolvable for forwarded misses */
-          if (strstr(filename, "libamdhip64"))
-            vgpu_real_hip("hipGetDeviceCount");
-          else if (strstr(filename, "libhsa-runtime64"))
-            vgpu_real_hsa("hsa_init");
-          else if (strstr(filename, "libamd_smi"))
-            vgpu_real_amdsmi("amdsmi_init");
-          else
-            vgpu_real_rsmi_handle();
+          /* the real library is resolved LAZILY on the first forwarded
+           * miss — an eager load here would force librocm_smi64 and
+           * libamd_smi into one process even when the app never calls
+           * into the other, re-creating the ODR crash the RTLD_NEXT
+           * resolution order avoids */
           return h;
         }
       }
@@ -195,8 +216,7 @@ void *dlsym(void *handle, const char *symbol) {
   if (!p && sym_v != NULL && g_self_handle && handle == g_self_handle) {
     tls_no_redirect++;
     if (strncmp(symbol, "rsmi_", 5) == 0) {
-      void *h = vgpu_real_rsmi_handle();
-      if (h) p = real(h, symbol);
+      p = vgpu_real_rsmi_sym(symbol);
     } else if (strncmp(symbol, "amdsmi_", 7) == 0) {
       p = vgpu_real_amdsmi(symbol);
     } else if (strncmp(symbol, "hsa_", 4) == 0) {

@@ -58,8 +58,7 @@ static int rsmi_index_to_vdev(uint32_t dv_ind) {
 }
 
 static void *real_rsmi_sym(const char *name) {
-  void *h = vgpu_real_rsmi_handle();
-  return h ? dlsym(h, name) : NULL;
+  return vgpu_real_rsmi_sym(name); /* RTLD_NEXT first — ODR-safe (hook.c) */
 }
 
 rsmi_status_t rsmi_dev_memory_total_get(uint32_t dv_ind,

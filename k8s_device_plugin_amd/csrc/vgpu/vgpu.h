@@ -132,6 +132,9 @@ void vgpu_limiter_gate(int dev, uint64_t workgroups); /* blocks when throttled *
 /* ---- hook core (hook.c) ---- */
 void *vgpu_real_hip(const char *sym);   /* resolve real libamdhip64 symbol */
 void *vgpu_real_rsmi_handle(void);
+/* ODR-safe rsmi resolution: RTLD_NEXT (in-process libamd_smi embedded
+ * copy) before dlopening librocm_smi64 — see hook.c */
+void *vgpu_real_rsmi_sym(const char *sym);
 void *vgpu_real_hsa(const char *sym);   /* resolve real libhsa-runtime64 symbol */
 void *vgpu_real_amdsmi(const char *sym); /* resolve real libamd_smi symbol */
 int vgpu_smi_index_to_vdev(uint32_t idx); /* smi.c: tool device idx -> vdev */

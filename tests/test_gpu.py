@@ -329,6 +329,24 @@ class TestLimiterReal:
 
 
 class TestSMISpoof:
+    def test_amdsmi_plt_link_under_preload(self, tmp_path):
+        """PLT-linked libamd_smi consumer under preload (the round-1
+        SIGBUS, ROUND2_NOTES item 5): root cause was dlopening
+        librocm_smi64 into a process whose libamd_smi embeds clashing
+        amd::smi C++ classes; the hook now resolves rsmi symbols via
+        RTLD_NEXT (libamd_smi's own embedded copy) so the process holds
+        ONE implementation.  The consumer must run and show the quota."""
+        out = subprocess.run(
+            [str(CSRC / "test" / "amdsmi_consumer")],
+            env={**os.environ, "LD_LIBRARY_PATH": "/opt/rocm/lib",
+                 **preload_env(tmp_path, limit="73728m")},
+            capture_output=True, text=True, timeout=120)
+        assert out.returncode == 0, \
+            f"rc={out.returncode} stderr: {out.stderr[-1500:]}"
+        rows = [json.loads(l) for l in out.stdout.splitlines()
+                if l.startswith("{")]
+        assert rows, out.stdout
+        assert rows[0]["total"] == 73728 * (1 << 20)
     def test_rsmi_reports_quota(self, tmp_path):
         """rocm-smi's library path: dlopen(librocm_smi64) is redirected and
         memory getters show the quota."""
