@@ -54,6 +54,10 @@ class MonitorCollector:
         ctr_desc = GaugeMetricFamily(
             "Device_memory_desc_of_container", "container memory breakdown",
             labels=["poduid", "ctrname", "vdeviceid", "deviceuuid", "bucket"])
+        ctr_scale = GaugeMetricFamily(
+            "vGPU_arbitrated_core_scale",
+            "node-arbitrated CU-throttle multiplier written to the region",
+            labels=["poduid", "ctrname", "vdeviceid", "deviceuuid"])
 
         for g in self.gpus:
             dev_dir = os.path.join(self.drm_root, f"card{g.drm_card}", "device")
@@ -92,8 +96,13 @@ class MonitorCollector:
                     buckets["data"] += buf_v
                 for name, val in buckets.items():
                     ctr_desc.add_metric(labels + [name], val)
+                try:
+                    ctr_scale.add_metric(
+                        labels, entry.region.get_monitor_scale(d))
+                except (OSError, ValueError):
+                    pass
 
-        return [host_mem, host_core, ctr_usage, ctr_limit, ctr_desc]
+        return [host_mem, host_core, ctr_usage, ctr_limit, ctr_desc, ctr_scale]
 
 
 def metrics_text(collector: MonitorCollector) -> bytes:
