@@ -84,3 +84,45 @@ amdsmi_status_t amdsmi_get_gpu_activity(amdsmi_processor_handle h,
   info->mm_activity = 0;
   return 0;
 }
+
+/* The REAL libamd_smi embeds and exports the whole rsmi_* surface; the
+ * interceptor must resolve rsmi symbols to THIS in-process copy via
+ * RTLD_NEXT instead of dlopening a second librocm_smi64 (whose C++
+ * statics clash -> the round-1 SIGBUS).  The counter proves the embedded
+ * copy was the one called. */
+static uint64_t g_embedded_rsmi_calls;
+
+uint64_t fake_amdsmi_embedded_rsmi_calls(void) {
+  return g_embedded_rsmi_calls;
+}
+
+typedef int rsmi_status_t;
+typedef int rsmi_memory_type_t;
+
+rsmi_status_t rsmi_dev_memory_total_get(uint32_t dv_ind,
+                                        rsmi_memory_type_t type,
+                                        uint64_t *total) {
+  (void)dv_ind;
+  if (type != 0 || !total) return 1;
+  g_embedded_rsmi_calls++;
+  *total = 288ULL << 30;
+  return 0;
+}
+
+rsmi_status_t rsmi_dev_memory_usage_get(uint32_t dv_ind,
+                                        rsmi_memory_type_t type,
+                                        uint64_t *used) {
+  (void)dv_ind;
+  if (type != 0 || !used) return 1;
+  g_embedded_rsmi_calls++;
+  *used = 200ULL << 30;
+  return 0;
+}
+
+rsmi_status_t rsmi_dev_busy_percent_get(uint32_t dv_ind, uint32_t *busy) {
+  (void)dv_ind;
+  if (!busy) return 1;
+  g_embedded_rsmi_calls++;
+  *busy = 90;
+  return 0;
+}

@@ -5,6 +5,7 @@
  * before dying (triage tooling for the real-library preload crash,
  * ROUND2_NOTES item 5). */
 #define _GNU_SOURCE
+#include <dlfcn.h>
 #include <execinfo.h>
 #include <signal.h>
 #include <stdint.h>
@@ -49,6 +50,8 @@ typedef struct {
 } amdsmi_engine_usage_t;
 extern amdsmi_status_t amdsmi_get_gpu_activity(amdsmi_processor_handle,
                                                amdsmi_engine_usage_t *);
+/* libamd_smi also exports the embedded rsmi surface */
+extern int rsmi_dev_memory_total_get(uint32_t, int, uint64_t *);
 
 int main(void) {
   if (getenv("AMDSMI_BT")) {
@@ -81,6 +84,22 @@ int main(void) {
            "\"vram_total_mb\":%u,\"vram_used_mb\":%u,\"gfx\":%u}\n",
            i, (unsigned long long)total, (unsigned long long)used,
            vu.vram_total, vu.vram_used, act.gfx_activity);
+  }
+  if (getenv("AMDSMI_CALL_RSMI")) {
+    /* exercise the embedded-rsmi interposition path (the one that
+     * SIGBUSed in round 1): a PLT call to rsmi_* from a libamd_smi
+     * consumer, resolved by the hook via RTLD_NEXT */
+    uint64_t total = 0;
+    int rc = rsmi_dev_memory_total_get(0, 0, &total);
+    long embedded = -1;
+    {
+      typedef unsigned long long (*cnt_fn)(void);
+      /* fake-lib introspection (absent on the real library) */
+      void *p = dlsym(RTLD_DEFAULT, "fake_amdsmi_embedded_rsmi_calls");
+      if (p) embedded = (long)((cnt_fn)p)();
+    }
+    printf("{\"rsmi_total\":%llu,\"rsmi_rc\":%d,\"embedded_calls\":%ld}\n",
+           (unsigned long long)total, rc, embedded);
   }
   fflush(stdout);
   return 0;

@@ -70,3 +70,20 @@ def test_amdsmi_activity_clamped_to_cu_limit(tmp_path):
     assert devs[0]["gfx"] == 25   # physical 90 clamped to quota
     devs = run_consumer(tmp_path / "r2.cache", limit="73728m")
     assert devs[0]["gfx"] == 90   # no CU limit -> passthrough
+
+
+def test_embedded_rsmi_resolved_via_rtld_next(tmp_path):
+    """ODR-safety regression test for the round-1 PLT-link SIGBUS: a
+    libamd_smi consumer's rsmi_* call is interposed by the hook, which
+    must resolve the REAL implementation from the in-process libamd_smi
+    (RTLD_NEXT — the fake's embedded-rsmi counter proves it) instead of
+    dlopening a second, clashing librocm_smi64."""
+    env_extra = {"AMDSMI_CALL_RSMI": "1"}
+    rows = run_consumer(tmp_path / "r.cache", limit="73728m",
+                        extra=env_extra)
+    rsmi_rows = [r for r in rows if "rsmi_total" in r]
+    assert rsmi_rows, rows
+    r = rsmi_rows[0]
+    assert r["rsmi_rc"] == 0
+    assert r["rsmi_total"] == 73728 * MIB      # quota view applied
+    assert r["embedded_calls"] >= 1            # served by the embedded copy
