@@ -43,20 +43,21 @@ def rows(pathmon: PathMonitor) -> List[dict]:
                 "state": ("BLOCKED" if snap.recent_kernel < 0 else
                           "active" if snap.recent_kernel > 0 else "idle"),
                 "oversub": bool(snap.oversubscribe),
+                "scale": e.region.get_monitor_scale(dev),
             })
     return out
 
 
 def render(rs: List[dict]) -> str:
     hdr = (f"{'POD':<38} {'CTR':<12} {'DEV':>3} {'USED':>9} {'LIMIT':>9} "
-           f"{'CU%':>4} {'PROCS':>5} {'PRIO':>4} {'STATE':<8} OS")
+           f"{'CU%':>4} {'SCALE':>6} {'PROCS':>5} {'PRIO':>4} {'STATE':<8} OS")
     lines = [hdr, "-" * len(hdr)]
     for r in rs:
         lines.append(
             f"{r['pod']:<38} {r['ctr']:<12} {r['dev']:>3} "
             f"{r['used_gib']:>8.1f}G {r['limit_gib']:>8.1f}G "
-            f"{r['cu_pct']:>4} {r['procs']:>5} {r['prio']:>4} "
-            f"{r['state']:<8} {'y' if r['oversub'] else '-'}")
+            f"{r['cu_pct']:>4} {r['scale']:>6.2f} {r['procs']:>5} "
+            f"{r['prio']:>4} {r['state']:<8} {'y' if r['oversub'] else '-'}")
     if not rs:
         lines.append("(no live vGPU containers)")
     return "\n".join(lines)
