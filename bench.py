@@ -53,6 +53,8 @@ def parse_args():
                    help="comma list or 'all' (the 10-case suite)")
     p.add_argument("--quota-pct", type=int, default=50)
     p.add_argument("--skip-colocated", action="store_true")
+    p.add_argument("--density-only", action="store_true",
+                   help="run just the density phase (fairness iteration)")
     p.add_argument("--density-pods", type=int, default=10,
                    help="pods in the density phase (0 disables; N=1 only)")
     p.add_argument("--density-seconds", type=float, default=20.0)
@@ -268,8 +270,22 @@ def run_density(args, base_env, tmp_root, total_mem):
     pods = args.density_pods
     quota_mib = total_mem // pods // (1 << 20)
     cu_pct = max(1, 100 // pods)
-    envs = [vgpu_env(base_env, tmp_root, quota_mib, f"density-{i}",
-                     cu_limit=cu_pct) for i in range(pods)]
+    envs = []
+    for i in range(pods):
+        e = vgpu_env(base_env, tmp_root, quota_mib, f"density-{i}",
+                     cu_limit=cu_pct)
+        # each pod gets its OWN MIOpen db copy: 10 processes racing one
+        # shared writable user-db can leave some pods on different conv
+        # solvers, skewing workgroups-per-sample and thus the fairness
+        # spread (token fairness equalizes workgroup rates)
+        for var in ("MIOPEN_USER_DB_PATH", "MIOPEN_CUSTOM_CACHE_DIR"):
+            src = e.get(var)
+            if src:
+                dst = os.path.join(tmp_root, f"{os.path.basename(src)}-p{i}")
+                if not os.path.isdir(dst):
+                    shutil.copytree(src, dst)
+                e[var] = dst
+        envs.append(e)
     caches = [e["VGPU_DEVICE_MEMORY_SHARED_CACHE"] for e in envs]
     cmd = [sys.executable, str(REPO / "bench.py"), "--density-worker",
            "--density-seconds", str(args.density_seconds)]
@@ -390,6 +406,15 @@ def main():
     total_mem = probe_total_mem(base_env)
     quota_bytes = total_mem * args.quota_pct // 100
     quota_mib = quota_bytes // (1 << 20)
+
+    if args.density_only:
+        density = run_density(args, base_env, tmp_root, total_mem)
+        shutil.rmtree(tmp_root, ignore_errors=True)
+        if rank == 0:
+            print(json.dumps({"density": density}), flush=True)
+        if dist is not None:
+            dist.destroy_process_group()
+        return
 
     t_job0 = time.perf_counter()
     native = run_phase(args, [base_env], barrier)[0]

@@ -34,11 +34,18 @@ def main():
                    help="working set as a multiple of physical HBM")
     p.add_argument("--chunk-gb", type=float, default=4.0)
     p.add_argument("--passes", type=int, default=3)
+    p.add_argument("--target-gb", type=float, default=0.0,
+                   help="absolute working set; use this under the preload, "
+                        "where total_memory reports the (larger) QUOTA and "
+                        "ratio-of-total would blow past the quota itself")
     args = p.parse_args()
 
     assert torch.cuda.is_available()
     hbm = torch.cuda.get_device_properties(0).total_memory
-    target = int(hbm * args.ratio)
+    target = int(args.target_gb * (1 << 30)) if args.target_gb \
+        else int(hbm * args.ratio)
+    # never exceed the allocator's own ceiling (quota) — leave 5% headroom
+    target = min(target, int(hbm * 0.95))
     chunk = int(args.chunk_gb * (1 << 30))
     n_chunks = max(1, target // chunk)
 

@@ -146,30 +146,51 @@ class TestOversubscriptionReal:
         under a 400 GB quota — allocations beyond the card must succeed
         (XNACK pages to host DRAM), data must survive, and the paging
         penalty is measured (chunk-fill time beyond physical vs in-HBM)."""
+        # Touch cost past physical is fault-bound (XNACK retries per page),
+        # so the beyond-physical region is touched in a BOUNDED 256 MB
+        # window per chunk — proving allocation + data integrity + a
+        # measured penalty without a multi-hundred-GB fault storm (the
+        # unbounded version blew a 15-minute budget).
+        # the quota view lies about total memory (that's the point), so
+        # learn the PHYSICAL size from an unpreloaded probe first
+        probe = run_child(
+            "import torch, json; torch.cuda.init();"
+            "print(json.dumps({'t':"
+            " torch.cuda.get_device_properties(0).total_memory}))", {})
+        phys_bytes = probe["t"]
         code = (
             "import torch, json, time; torch.cuda.init()\n"
-            "phys = torch.cuda.get_device_properties(0).total_memory\n"
+            f"phys = {phys_bytes}\n"
             "chunk = 8 << 30\n"
-            "n = (phys + (16 << 30)) // chunk + 1   # ~16 GB past physical\n"
-            "chunks, times = [], []\n"
-            "for i in range(int(n)):\n"
-            "    x = torch.empty(chunk, dtype=torch.uint8, device='cuda')\n"
-            "    t0 = time.perf_counter(); x.fill_(7)\n"
+            "def map_pages(x, val):\n"
+            "    t0 = time.perf_counter()\n"
+            "    x[::4096] = val  # one write per 4K page: backs the chunk\n"
             "    torch.cuda.synchronize()\n"
-            "    times.append(time.perf_counter() - t0)\n"
+            "    return time.perf_counter() - t0\n"
+            "n_resident = max(0, int((phys - (8 << 30)) // chunk))\n"
+            "chunks, t_res = [], []\n"
+            "for i in range(n_resident):\n"
+            "    x = torch.empty(chunk, dtype=torch.uint8, device='cuda')\n"
+            "    t_res.append(map_pages(x, 7))\n"
+            "    chunks.append(x)\n"
+            "t_over = []\n"
+            "for i in range(3):  # 24 GB more: crosses the 288 GB card\n"
+            "    x = torch.empty(chunk, dtype=torch.uint8, device='cuda')\n"
+            "    t_over.append(map_pages(x, 9))\n"
             "    chunks.append(x)\n"
             "held = len(chunks) * chunk\n"
-            "ok = int(chunks[0][:64].sum()) == 64 * 7 and \\\n"
-            "     int(chunks[-1][:64].sum()) == 64 * 7\n"
-            "penalty = max(times[-3:]) / max(min(times[1:4]), 1e-9)\n"
+            "ok = int(chunks[0][0]) == 7 and int(chunks[-1][0]) == 9 and \\\n"
+            "     int(chunks[n_resident - 1][4096]) == 7\n"
             "print(json.dumps({'held_gb': held >> 30,"
             " 'phys_gb': phys >> 30, 'ok': ok,"
-            " 'penalty_x': round(penalty, 1)}))"
+            " 'penalty_x': round(max(t_over) / max(min(t_res), 1e-9), 1),"
+            " 't_res_s': round(sum(t_res), 1),"
+            " 't_over_s': round(sum(t_over), 1)}))"
         )
         res = run_child(code, preload_env(
             tmp_path, limit="409600m",
             extra={"VGPU_OVERSUBSCRIBE": "true", "HSA_XNACK": "1"}),
-            timeout=900)
+            timeout=600)
         assert res["held_gb"] > res["phys_gb"], \
             "never exceeded physical HBM"
         assert res["ok"], "data corrupted across the paging boundary"
