@@ -89,3 +89,27 @@ def test_gloo_world2_sum_and_max():
     # every rank sees the same aggregates: 1000+2000 and max(10,15)
     assert results[0] == (3000.0, 15.0)
     assert results[1] == (3000.0, 15.0)
+
+
+@pytest.mark.timeout(180)
+def test_bench_density_only_fake():
+    """Density-phase choreography (10-worker spawn, READY/GO barrier,
+    arbitration thread, fairness JSON) on CPU via BENCH_FAKE_GPU."""
+    import json
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    env = dict(os.environ)
+    env["BENCH_FAKE_GPU"] = "1"
+    out = subprocess.run(
+        [sys.executable, str(repo / "bench.py"), "--density-only",
+         "--density-pods", "3", "--density-seconds", "0.2"],
+        env=env, capture_output=True, text=True, timeout=160, cwd=str(repo))
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][0]
+    res = json.loads(line)
+    assert res["density"]["pods"] == 3
+    assert len(res["density"]["per_pod_samples_per_s"]) == 3
+    assert res["density"]["fairness_max_over_min"] >= 1.0
