@@ -356,11 +356,25 @@ def run_density(args, base_env, tmp_root, total_mem):
             if proc.poll() is None:
                 proc.kill()
     rates = [r["samples_per_s"] for r in results]
+    # limiter introspection per pod: which control branch actually ran
+    debug = []
+    for c in caches:
+        try:
+            r = SharedRegion(c)
+            debug.append({
+                "fill_rate": r.get_token_fill_rate(0),
+                "scale": round(r.get_monitor_scale(0), 3),
+                "sm_limit": r.snapshot().sm_limit[0],
+            })
+            r.close()
+        except (OSError, ValueError, IndexError):
+            debug.append(None)
     return {
         "pods": pods,
         "aggregate_samples_per_s": round(sum(rates), 2),
         "per_pod_samples_per_s": [round(r, 2) for r in rates],
         "fairness_max_over_min": round(max(rates) / max(min(rates), 1e-9), 3),
+        "limiter_debug": debug,
     }
 
 

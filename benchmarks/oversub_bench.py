@@ -49,11 +49,22 @@ def main():
     chunk = int(args.chunk_gb * (1 << 30))
     n_chunks = max(1, target // chunk)
 
+    # bulk-migrate each chunk into HBM with hipMemPrefetchAsync: XNACK
+    # demand faults back fresh pages one retry at a time (minutes per GB),
+    # while prefetch DMA-moves whole ranges at engine speed — the
+    # MI355X-native way to establish residency
+    import ctypes
+    hip = ctypes.CDLL("libamdhip64.so")
+    hip.hipMemPrefetchAsync.argtypes = [
+        ctypes.c_void_p, ctypes.c_size_t, ctypes.c_int, ctypes.c_void_p]
+
     chunks = []
     t0 = time.perf_counter()
     for i in range(n_chunks):
-        chunks.append(torch.zeros(chunk // 4, dtype=torch.float32,
-                                  device="cuda"))
+        x = torch.empty(chunk // 4, dtype=torch.float32, device="cuda")
+        hip.hipMemPrefetchAsync(x.data_ptr(), chunk, 0, None)
+        torch.cuda.synchronize()
+        chunks.append(x)
     torch.cuda.synchronize()
     alloc_s = time.perf_counter() - t0
 

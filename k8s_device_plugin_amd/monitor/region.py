@@ -135,6 +135,18 @@ class SharedRegion:
         return struct.unpack_from(
             "<Q", self._mm, self.layout["monitor_interval_ns"])[0] / 1e9
 
+    # -- limiter introspection (debug/observability) ------------------------
+    def get_token_fill_rate(self, dev: int) -> int:
+        """tokens/s the limiter last applied (post-scale) — written by
+        refill(); reveals which control branch (arbitrated vs local EMA)
+        a container is actually running under."""
+        return struct.unpack_from(
+            "<q", self._mm, self.layout["token_fill_rate"] + 8 * dev)[0]
+
+    def get_core_tokens(self, dev: int) -> int:
+        return struct.unpack_from(
+            "<q", self._mm, self.layout["core_tokens"] + 8 * dev)[0]
+
     def get_monitor_scale(self, dev: int) -> float:
         v = struct.unpack_from(
             "<q", self._mm, self.layout["monitor_scale_fp"] + 8 * dev)[0]
