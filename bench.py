@@ -293,20 +293,17 @@ def run_density(args, base_env, tmp_root, total_mem):
                               stdout=subprocess.PIPE, text=True,
                               cwd=str(REPO)) for e in envs]
 
-    # monitor-style arbitration: one AIMD scale for the device, written to
-    # every pod's region (monitor/feedback.py _arbitrate, in-process here
-    # because there is no monitor daemon on a bench box)
+    # monitor-style arbitration (monitor/feedback.py _arbitrate, in-process
+    # because there is no monitor daemon on a bench box): tighten the
+    # common scale until every pod is marginally token-bound, relax slowly
+    # once all are — max-min fairness from the buckets alone (host busy%
+    # proved misleading on multi-DRM hosts; profiles/r02_summary.md)
     stop = threading.Event()
-
-    def busy_path():
-        import glob as g
-        cards = sorted(g.glob("/sys/class/drm/card*/device/gpu_busy_percent"))
-        return cards[0] if cards else None
 
     def arbitrate():
         regions = {}
+        prev_tokens = {}
         scale = 1.0
-        path = busy_path()
         while not stop.is_set():
             for c in caches:
                 if c not in regions:
@@ -316,16 +313,27 @@ def run_density(args, base_env, tmp_root, total_mem):
                             regions[c] = r
                     except (OSError, ValueError):
                         pass
-            busy = -1
-            if path:
+            active = 0
+            bound = 0
+            for c, r in regions.items():
                 try:
-                    busy = int(open(path).read().strip())
+                    tokens = r.get_core_tokens(0)
+                    fill = r.get_token_fill_rate(0)
                 except (OSError, ValueError):
-                    pass
-            if busy > 95:
-                scale = max(0.05, scale * 0.90)
-            elif 0 <= busy < 85:
-                scale = min(100.0, scale * 1.10)
+                    continue
+                cap = max(1.0, fill * 0.25)
+                moved = prev_tokens.get(c) != tokens
+                prev_tokens[c] = tokens
+                if tokens <= 0:
+                    active += 1
+                    bound += 1
+                elif moved or tokens < 0.9 * cap:
+                    active += 1  # launching but not (yet) bound
+            if active:
+                # median-bound target: clip the heavy users, let light
+                # users run free (see monitor/feedback.py _arbitrate)
+                scale = (min(100.0, scale * 1.02) if 2 * bound >= active
+                         else max(0.05, scale * 0.95))
             now = time.monotonic_ns()
             for r in regions.values():
                 try:

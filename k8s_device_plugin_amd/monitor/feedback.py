@@ -27,7 +27,7 @@ ACTIVITY_THRESHOLD = 0  # recent_kernel above this => active
 # co-located containers share ONE multiplier on their entitled CU share —
 # proportional fairness without per-process attribution)
 SCALE_LO, SCALE_HI = 0.05, 100.0
-BUSY_HIGH, BUSY_LOW = 95, 85
+TIGHTEN, RELAX = 0.95, 1.02
 
 
 class FeedbackLoop:
@@ -37,15 +37,18 @@ class FeedbackLoop:
         """soft_cores=True: enforce the CU limit only while the device is
         contended (the reference's default GPU_CORE_UTILIZATION_POLICY);
         False (our default): strict isolation, limit always enforced.
-        busy_reader(uuid) -> device busy percent enables node-arbitrated
-        fair throttling (monitor writes one scale to every region).
-        interval_s is this loop's period; it is published into every
-        region so the limiter sizes its scale-freshness window to it."""
+        busy_reader(uuid) -> device busy percent is kept for metrics and
+        backward compatibility; the arbitration controller itself runs on
+        the regions' token buckets (see _arbitrate) and needs no host
+        utilization signal.  interval_s is this loop's period; it is
+        published into every region so the limiter sizes its
+        scale-freshness window to it."""
         self.pathmon = pathmon
         self.soft_cores = soft_cores
         self.busy_reader = busy_reader
         self.interval_s = interval_s
         self._scale: Dict[str, float] = {}  # device uuid -> multiplier
+        self._activity: Dict[str, bool] = {}
 
     def observe_once(self) -> None:
         entries = self.pathmon.live_regions()
@@ -57,6 +60,7 @@ class FeedbackLoop:
             rk = e.region.get_recent_kernel()
             active = rk > ACTIVITY_THRESHOLD and bool(snap.procs)
             activity[e.key] = active
+            self._activity = activity
             # decay: an idle container's recent_kernel drifts to 0
             if rk > 0:
                 e.region.set_recent_kernel(rk - 1)
@@ -99,36 +103,57 @@ class FeedbackLoop:
             else:
                 e.region.set_utilization_switch(1)
 
-        if self.busy_reader is not None:
-            self._arbitrate(by_device)
+        self._arbitrate(by_device)
 
     def _arbitrate(self, by_device) -> None:
-        """One AIMD-style scale per device, written to every region holding
-        it: the limiter multiplies each container's entitled share by this
-        common factor, so shares stay proportional while total device busy
-        converges just under saturation."""
+        """One scale per device, written to every region holding it.
+
+        Controller: drive the common multiplier toward the point where
+        HALF the active CU-limited containers are TOKEN-BOUND (bucket
+        sampled at/below zero): tighten (x0.95) while fewer than half
+        bind, relax slowly (x1.02) otherwise.  The high-demand pods bind
+        first as the scale drops, so equilibrium clips the heavy users at
+        a common token rate near the median demand while light users run
+        free below it — single-knob max-min fairness over the entitled
+        shares, using only state the monitor can actually observe (the
+        buckets in the regions).  Requiring ALL pods to bind would let
+        one low-demand pod drag the scale to the floor; a host busy%
+        signal is NOT needed and was actively misleading: on multi-DRM
+        hosts the wrong card reads ~0 busy and an AIMD-on-busy controller
+        saturates the scale, unthrottling everyone (measured: fairness
+        19x r1, 2.1x with the busy controller; profiles/r02_summary.md)."""
         now = time.monotonic_ns()
         for uuid, prio_map in by_device.items():
-            try:
-                busy = self.busy_reader(uuid)
-            except Exception as e:
-                log.warning("busy read for %s failed: %s", uuid, e)
-                continue
-            if busy is None or busy < 0:
-                continue
+            ents = [e for es in prio_map.values() for e in es]
+            active = 0
+            bound = 0
+            targets = []
+            for e in ents:
+                try:
+                    snap = e.region.snapshot()
+                    dev = snap.uuids.index(uuid)
+                except (ValueError, OSError) as exc:
+                    log.debug("region read %s failed: %s", e.key, exc)
+                    continue
+                targets.append((e, dev))
+                lim = snap.sm_limit[dev]
+                if not (0 < lim < 100):
+                    continue
+                if not self._activity.get(e.key):
+                    continue
+                active += 1
+                try:
+                    if e.region.get_core_tokens(dev) <= 0:
+                        bound += 1
+                except (OSError, ValueError):
+                    pass
             scale = self._scale.get(uuid, 1.0)
-            if busy > BUSY_HIGH:
-                scale *= 0.90
-            elif busy < BUSY_LOW:
-                scale *= 1.10
-            scale = max(SCALE_LO, min(SCALE_HI, scale))
-            self._scale[uuid] = scale
-            for ents in prio_map.values():
-                for e in ents:
-                    try:
-                        snap = e.region.snapshot()
-                        for dev, u in enumerate(snap.uuids):
-                            if u == uuid:
-                                e.region.set_monitor_scale(dev, scale, now)
-                    except Exception as exc:
-                        log.warning("scale write %s failed: %s", e.key, exc)
+            if active:
+                scale *= RELAX if 2 * bound >= active else TIGHTEN
+                scale = max(SCALE_LO, min(SCALE_HI, scale))
+                self._scale[uuid] = scale
+            for e, dev in targets:
+                try:
+                    e.region.set_monitor_scale(dev, scale, now)
+                except Exception as exc:
+                    log.warning("scale write %s failed: %s", e.key, exc)
