@@ -156,7 +156,13 @@ def main():
         arb_stop = threading.Event()
 
         def arbitrate():
+            # token-bound median controller (monitor/feedback.py
+            # _arbitrate): tighten while fewer than half the launching
+            # pods sample token-bound, relax slowly otherwise — fairness
+            # from the buckets alone, no host busy% (which reads the
+            # wrong card on multi-DRM hosts and unthrottles everyone)
             regions = {}
+            prev_tokens = {}
             scale = 1.0
             while not arb_stop.is_set():
                 # late-attach: the workers create/initialize their regions
@@ -170,16 +176,25 @@ def main():
                             regions[c] = r
                     except (OSError, ValueError):
                         pass
-                busy = -1
-                if busy_path:
+                active = 0
+                bound = 0
+                for c, r in regions.items():
                     try:
-                        busy = int(open(busy_path).read().strip())
+                        tokens = r.get_core_tokens(0)
+                        fill = r.get_token_fill_rate(0)
                     except (OSError, ValueError):
-                        pass
-                if busy > 95:
-                    scale = max(0.05, scale * 0.90)
-                elif 0 <= busy < 85:
-                    scale = min(100.0, scale * 1.10)
+                        continue
+                    cap = max(1.0, fill * 0.25)
+                    moved = prev_tokens.get(c) != tokens
+                    prev_tokens[c] = tokens
+                    if tokens <= 0:
+                        active += 1
+                        bound += 1
+                    elif moved or tokens < 0.9 * cap:
+                        active += 1
+                if active:
+                    scale = (min(100.0, scale * 1.02) if 2 * bound >= active
+                             else max(0.05, scale * 0.95))
                 now = time.monotonic_ns()
                 for r in regions.values():
                     try:
