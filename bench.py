@@ -301,8 +301,11 @@ def run_density(args, base_env, tmp_root, total_mem):
     stop = threading.Event()
 
     def arbitrate():
+        from k8s_device_plugin_amd.monitor.arbiter import ScaleArbiter
+
         regions = {}
         prev_tokens = {}
+        arb = ScaleArbiter()
         scale = 1.0
         while not stop.is_set():
             for c in caches:
@@ -329,11 +332,7 @@ def run_density(args, base_env, tmp_root, total_mem):
                     bound += 1
                 elif moved or tokens < 0.9 * cap:
                     active += 1  # launching but not (yet) bound
-            if active:
-                # median-bound target: clip the heavy users, let light
-                # users run free (see monitor/feedback.py _arbitrate)
-                scale = (min(100.0, scale * 1.02) if 2 * bound >= active
-                         else max(0.05, scale * 0.95))
+            scale = arb.tick(active, bound)
             now = time.monotonic_ns()
             for r in regions.values():
                 try:

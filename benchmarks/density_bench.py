@@ -161,8 +161,11 @@ def main():
             # pods sample token-bound, relax slowly otherwise — fairness
             # from the buckets alone, no host busy% (which reads the
             # wrong card on multi-DRM hosts and unthrottles everyone)
+            from k8s_device_plugin_amd.monitor.arbiter import ScaleArbiter
+
             regions = {}
             prev_tokens = {}
+            arb = ScaleArbiter()
             scale = 1.0
             while not arb_stop.is_set():
                 # late-attach: the workers create/initialize their regions
@@ -192,9 +195,7 @@ def main():
                         bound += 1
                     elif moved or tokens < 0.9 * cap:
                         active += 1
-                if active:
-                    scale = (min(100.0, scale * 1.02) if 2 * bound >= active
-                             else max(0.05, scale * 0.95))
+                scale = arb.tick(active, bound)
                 now = time.monotonic_ns()
                 for r in regions.values():
                     try:

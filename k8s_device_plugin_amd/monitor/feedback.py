@@ -17,18 +17,12 @@ import logging
 import time
 from typing import Callable, Dict, List, Optional, Set, Tuple
 
+from .arbiter import ScaleArbiter
 from .pathmon import ContainerEntry, PathMonitor
 
 log = logging.getLogger(__name__)
 
 ACTIVITY_THRESHOLD = 0  # recent_kernel above this => active
-
-# node-arbitrated throttle scale bounds (written into every region so all
-# co-located containers share ONE multiplier on their entitled CU share —
-# proportional fairness without per-process attribution)
-SCALE_LO, SCALE_HI = 0.05, 100.0
-TIGHTEN, RELAX = 0.95, 1.02
-
 
 class FeedbackLoop:
     def __init__(self, pathmon: PathMonitor, soft_cores: bool = False,
@@ -47,7 +41,8 @@ class FeedbackLoop:
         self.soft_cores = soft_cores
         self.busy_reader = busy_reader
         self.interval_s = interval_s
-        self._scale: Dict[str, float] = {}  # device uuid -> multiplier
+        # device uuid -> slow-start controller (see monitor/arbiter.py)
+        self._arbiters: Dict[str, ScaleArbiter] = {}
         self._activity: Dict[str, bool] = {}
 
     def observe_once(self) -> None:
@@ -108,20 +103,10 @@ class FeedbackLoop:
     def _arbitrate(self, by_device) -> None:
         """One scale per device, written to every region holding it.
 
-        Controller: drive the common multiplier toward the point where
-        HALF the active CU-limited containers are TOKEN-BOUND (bucket
-        sampled at/below zero): tighten (x0.95) while fewer than half
-        bind, relax slowly (x1.02) otherwise.  The high-demand pods bind
-        first as the scale drops, so equilibrium clips the heavy users at
-        a common token rate near the median demand while light users run
-        free below it — single-knob max-min fairness over the entitled
-        shares, using only state the monitor can actually observe (the
-        buckets in the regions).  Requiring ALL pods to bind would let
-        one low-demand pod drag the scale to the floor; a host busy%
-        signal is NOT needed and was actively misleading: on multi-DRM
-        hosts the wrong card reads ~0 busy and an AIMD-on-busy controller
-        saturates the scale, unthrottling everyone (measured: fairness
-        19x r1, 2.1x with the busy controller; profiles/r02_summary.md)."""
+        Control law and dynamics: monitor/arbiter.py (token-bound
+        median target with slow-start).  This method only gathers the
+        (active, bound) observation per device and publishes the scale
+        to every region holding it."""
         now = time.monotonic_ns()
         for uuid, prio_map in by_device.items():
             ents = [e for es in prio_map.values() for e in es]
@@ -147,11 +132,8 @@ class FeedbackLoop:
                         bound += 1
                 except (OSError, ValueError):
                     pass
-            scale = self._scale.get(uuid, 1.0)
-            if active:
-                scale *= RELAX if 2 * bound >= active else TIGHTEN
-                scale = max(SCALE_LO, min(SCALE_HI, scale))
-                self._scale[uuid] = scale
+            arb = self._arbiters.setdefault(uuid, ScaleArbiter())
+            scale = arb.tick(active, bound)
             for e, dev in targets:
                 try:
                     e.region.set_monitor_scale(dev, scale, now)

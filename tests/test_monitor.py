@@ -293,7 +293,7 @@ class TestArbitratedThrottle:
                 fb.observe_once()  # keep the ts fresh
                 for e in pathmon.live_regions():
                     e.region.set_recent_kernel(5)
-                expect = 4_000_000 * 0.10 * fb._scale["GPU-test-0"]
+                expect = 4_000_000 * 0.10 * fb._arbiters["GPU-test-0"].scale
                 rate = _struct.unpack_from(
                     "<q", region._mm, region.layout["token_fill_rate"])[0]
                 if rate and abs(rate - expect) / expect < 0.30:
@@ -303,7 +303,7 @@ class TestArbitratedThrottle:
             assert ok, f"limiter did not adopt monitor scale (rate={rate}, expect={expect})"
 
             # once EVERY active pod is token-bound the controller relaxes
-            before = fb._scale["GPU-test-0"]
+            before = fb._arbiters["GPU-test-0"].scale
             for c in caches:
                 r = SharedRegion(str(c))
                 _struct.pack_into("<q", r._mm, r.layout["core_tokens"], -5)
@@ -311,7 +311,7 @@ class TestArbitratedThrottle:
             for e in pathmon.live_regions():
                 e.region.set_recent_kernel(5)
             fb.observe_once()
-            assert fb._scale["GPU-test-0"] > before
+            assert fb._arbiters["GPU-test-0"].scale > before
         finally:
             for p in procs:
                 p.kill()
