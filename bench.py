@@ -365,13 +365,25 @@ def run_density(args, base_env, tmp_root, total_mem):
     rates = [r["samples_per_s"] for r in results]
     # limiter introspection per pod: which control branch actually ran
     debug = []
+    now_ns = time.monotonic_ns()
     for c in caches:
         try:
             r = SharedRegion(c)
+            snap = r.snapshot()
+            L = r.layout
+            import struct as _st
+            ts = _st.unpack_from("<Q", r._mm, L["monitor_scale_ts_ns"])[0]
+            last_refill = _st.unpack_from("<Q", r._mm, L["last_refill_ns"])[0]
             debug.append({
                 "fill_rate": r.get_token_fill_rate(0),
+                "tokens": r.get_core_tokens(0),
                 "scale": round(r.get_monitor_scale(0), 3),
-                "sm_limit": r.snapshot().sm_limit[0],
+                "sm_limit": snap.sm_limit[0],
+                "switch": snap.utilization_switch,
+                "recent_kernel": snap.recent_kernel,
+                "scale_ts_age_s": round((now_ns - ts) / 1e9, 2) if ts else -1,
+                "refill_age_s": round((now_ns - last_refill) / 1e9, 2)
+                                if last_refill else -1,
             })
             r.close()
         except (OSError, ValueError, IndexError):
