@@ -470,6 +470,17 @@ void vgpu_limiter_gate(int dev, uint64_t workgroups) {
 /* ---- launch hooks ---------------------------------------------------- */
 typedef struct { unsigned x, y, z; } vdim3;
 
+/* Launch cost in WAVEFRONTS (the CDNA4 issue unit: 64-wide), not bare
+ * workgroups: co-located pods running different conv solvers can differ
+ * 2x in threads-per-workgroup, and workgroup-charging hands the
+ * fat-workgroup solver a 2x samples/s advantage at equal token rates
+ * (measured: the single 433-vs-217 outlier pod in the density runs). */
+static inline uint64_t launch_cost(uint64_t groups, uint64_t threads_per_group) {
+  uint64_t waves = (threads_per_group + 63) / 64;
+  if (waves < 1) waves = 1;
+  return groups * waves;
+}
+
 /* ---- graph workgroup accounting ---------------------------------------
  * A replayed graph bundles many kernel launches and the per-node hooks do
  * not fire on replay, so hipGraphLaunch must charge the graph's REAL
@@ -517,7 +528,11 @@ static uint64_t graph_workgroups(void *graph) {
         uint64_t g = (uint64_t)(p.gridDim.x ? p.gridDim.x : 1) *
                      (p.gridDim.y ? p.gridDim.y : 1) *
                      (p.gridDim.z ? p.gridDim.z : 1);
-        cost += g;
+        uint64_t t = (uint64_t)(p.blockDim.x ? p.blockDim.x : 1) *
+                     (p.blockDim.y ? p.blockDim.y : 1) *
+                     (p.blockDim.z ? p.blockDim.z : 1);
+        uint64_t waves = (t + 63) / 64;
+        cost += g * (waves ? waves : 1);
       }
     }
   }
@@ -614,7 +629,8 @@ hipError_t hipLaunchKernel(const void *f, vdim3 grid, vdim3 block, void **args,
   vgpu_ensure_initialized();
   if (!vgpu_control_disabled())
     vgpu_limiter_gate(vgpu_current_device(),
-                      (uint64_t)grid.x * grid.y * grid.z);
+                      launch_cost((uint64_t)grid.x * grid.y * grid.z,
+                                  (uint64_t)block.x * block.y * block.z));
   return real(f, grid, block, args, shared, stream);
 }
 
@@ -630,7 +646,9 @@ hipError_t hipModuleLaunchKernel(void *func, unsigned gx, unsigned gy,
   if (!real) return hipErrorInvalidValue;
   vgpu_ensure_initialized();
   if (!vgpu_control_disabled())
-    vgpu_limiter_gate(vgpu_current_device(), (uint64_t)gx * gy * gz);
+    vgpu_limiter_gate(vgpu_current_device(),
+                      launch_cost((uint64_t)gx * gy * gz,
+                                  (uint64_t)bx * by * bz));
   return real(func, gx, gy, gz, bx, by, bz, sharedMem, stream, params, extra);
 }
 
@@ -651,7 +669,8 @@ hipError_t hipExtModuleLaunchKernel(void *func, unsigned gwx, unsigned gwy,
     uint64_t wgs = ((uint64_t)gwx / (bx ? bx : 1)) *
                    ((uint64_t)gwy / (by ? by : 1)) *
                    ((uint64_t)gwz / (bz ? bz : 1));
-    vgpu_limiter_gate(vgpu_current_device(), wgs ? wgs : 1);
+    vgpu_limiter_gate(vgpu_current_device(),
+                      launch_cost(wgs ? wgs : 1, (uint64_t)bx * by * bz));
   }
   return real(func, gwx, gwy, gwz, bx, by, bz, sharedMem, stream, startEvent,
               stopEvent, flags);
@@ -668,7 +687,8 @@ hipError_t hipLaunchCooperativeKernel(const void *f, vdim3 grid, vdim3 block,
   vgpu_ensure_initialized();
   if (!vgpu_control_disabled())
     vgpu_limiter_gate(vgpu_current_device(),
-                      (uint64_t)grid.x * grid.y * grid.z);
+                      launch_cost((uint64_t)grid.x * grid.y * grid.z,
+                                  (uint64_t)block.x * block.y * block.z));
   return real(f, grid, block, args, shared, stream);
 }
 
@@ -700,7 +720,8 @@ hipError_t hipExtLaunchKernel(const void *f, vdim3 grid, vdim3 block,
   vgpu_ensure_initialized();
   if (!vgpu_control_disabled())
     vgpu_limiter_gate(vgpu_current_device(),
-                      (uint64_t)grid.x * grid.y * grid.z);
+                      launch_cost((uint64_t)grid.x * grid.y * grid.z,
+                                  (uint64_t)block.x * block.y * block.z));
   return real(f, grid, block, args, shared, stream, startEvent, stopEvent,
               flags);
 }
