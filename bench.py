@@ -58,6 +58,10 @@ def parse_args():
     p.add_argument("--density-pods", type=int, default=10,
                    help="pods in the density phase (0 disables; N=1 only)")
     p.add_argument("--density-seconds", type=float, default=20.0)
+    p.add_argument("--density-settle", type=float, default=6.0,
+                   help="untimed run-in before the timed density window so "
+                        "the arbiter reaches steady state (the slow-start "
+                        "transient otherwise lands inside the measurement)")
     p.add_argument("--worker", action="store_true", help=argparse.SUPPRESS)
     p.add_argument("--density-worker", action="store_true",
                    help=argparse.SUPPRESS)
@@ -77,11 +81,9 @@ def miopen_env(tmp_root):
     """Writable copies of the shipped pre-tuned MIOpen dbs (if present).
 
     MIOPEN_FIND_MODE=FAST makes solver selection deterministic: find-db
-    hit, else the immediate-mode heuristic — never a runtime re-search.
-    Ten co-located pods auto-tuning concurrently pick different winners
-    under each other's noise, and a pod whose solver needs half the
-    workgroups per sample runs 2x faster under workgroup-fair throttling
-    (measured: the single 600-samples/s outlier in the density runs)."""
+    hit, else the immediate-mode heuristic — never a runtime re-search,
+    so co-located pods cannot diverge by auto-tuning under each other's
+    noise and every phase compares identical kernel streams."""
     env = {"MIOPEN_FIND_MODE": "FAST"}
     if MIOPEN_UDB.is_dir():
         dst = os.path.join(tmp_root, "miopen_udb")
@@ -188,6 +190,13 @@ def density_worker_main(args):
     torch.cuda.synchronize()
     print("READY", flush=True)
     assert sys.stdin.readline().strip() == "GO"
+    # settle: run untimed while the arbiter's slow-start converges, so the
+    # timed window samples steady state (the coarse-phase overshoot lets
+    # whichever queue KFD favors bank a free-run lead otherwise)
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < args.density_settle:
+        zoo.step(case, model, batch, None)
+    torch.cuda.synchronize()
     t0 = time.perf_counter()
     steps = 0
     while time.perf_counter() - t0 < args.density_seconds:
@@ -295,7 +304,8 @@ def run_density(args, base_env, tmp_root, total_mem):
         envs.append(e)
     caches = [e["VGPU_DEVICE_MEMORY_SHARED_CACHE"] for e in envs]
     cmd = [sys.executable, str(REPO / "bench.py"), "--density-worker",
-           "--density-seconds", str(args.density_seconds)]
+           "--density-seconds", str(args.density_seconds),
+           "--density-settle", str(args.density_settle)]
     procs = [subprocess.Popen(cmd, env=e, stdin=subprocess.PIPE,
                               stdout=subprocess.PIPE, text=True,
                               cwd=str(REPO)) for e in envs]

@@ -109,6 +109,21 @@ int main(int argc, char **argv) {
           "{\"cmd\":\"launch\",\"n\":%ld,\"grid\":%u,\"seconds\":%.6f,"
           "\"err\":%d}\n",
           n, grid, dt, e);
+    } else if (strcmp(cmd, "launchb") == 0 && i + 3 < argc) {
+      /* like launch, with an explicit block size (wavefront-cost tests) */
+      long n = atol(argv[++i]);
+      unsigned grid = (unsigned)atoi(argv[++i]);
+      unsigned block = (unsigned)atoi(argv[++i]);
+      vdim3 g = {grid, 1, 1}, b = {block, 1, 1};
+      double t0 = now_s();
+      hipError_t e = 0;
+      for (long k = 0; k < n && e == 0; k++)
+        e = hipLaunchKernel((void *)main, g, b, NULL, 0, NULL);
+      double dt = now_s() - t0;
+      printf(
+          "{\"cmd\":\"launchb\",\"n\":%ld,\"grid\":%u,\"block\":%u,"
+          "\"seconds\":%.6f,\"err\":%d}\n",
+          n, grid, block, dt, e);
     } else if (strcmp(cmd, "graphlaunch") == 0 && i + 3 < argc) {
       long launches = atol(argv[++i]);
       long nodes = atol(argv[++i]);

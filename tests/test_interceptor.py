@@ -196,6 +196,28 @@ class TestLimiter:
         )
         assert res[0]["seconds"] < 0.2
 
+    def test_wavefront_cost_scales_with_block_size(self, tmp_path):
+        """Launch cost is wavefronts (grid x ceil(block/64)), the CDNA4
+        issue unit: a 256-thread block costs 4x a 64-thread block, so a
+        fat-workgroup solver cannot out-run a thin one at equal token
+        rates.  100 x 1250 wg x 4 waves = 500k tokens @1M/s >= ~0.25 s;
+        the same workgroup count at block 64 is 125k tokens -> fast."""
+        res = run_consumer(
+            ["launchb", 100, 1250, 256], tmp_path / "a.cache",
+            extra_env={"VGPU_DEVICE_CU_LIMIT": "10",
+                       "VGPU_TOKEN_RATE": "1000000"},
+        )
+        assert res[0]["err"] == 0
+        assert res[0]["seconds"] > 0.2, \
+            f"fat blocks under-charged: {res[0]['seconds']}"
+        res2 = run_consumer(
+            ["launchb", 100, 1250, 64], tmp_path / "b.cache",
+            extra_env={"VGPU_DEVICE_CU_LIMIT": "10",
+                       "VGPU_TOKEN_RATE": "1000000"},
+        )
+        assert res2[0]["seconds"] < 0.1, \
+            f"thin blocks over-charged: {res2[0]['seconds']}"
+
     def test_limit_100_not_throttled(self, tmp_path):
         res = run_consumer(
             ["launch", 100, 5000], tmp_path / "r.cache",
