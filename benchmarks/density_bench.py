@@ -46,6 +46,10 @@ def worker(args):
     print("READY", flush=True)
     assert sys.stdin.readline().strip() == "GO"
     t0 = time.perf_counter()
+    while time.perf_counter() - t0 < args.settle:
+        zoo.step(case, model, batch, None)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
     steps = 0
     while time.perf_counter() - t0 < args.seconds:
         zoo.step(case, model, batch, None)
@@ -60,6 +64,9 @@ def main():
     p = argparse.ArgumentParser()
     p.add_argument("--pods", type=int, default=10)
     p.add_argument("--seconds", type=float, default=30.0)
+    p.add_argument("--settle", type=float, default=6.0,
+                   help="untimed run-in before the timed window (arbiter "
+                        "steady state)")
     p.add_argument("--case", default="resnet50_inf")
     p.add_argument("--quota-pct", type=int, default=0,
                    help="HBM percent per pod (default 100/pods)")
@@ -136,7 +143,7 @@ def main():
             env["HSA_CU_MASK"] = masks[i]
         procs.append(subprocess.Popen(
             [sys.executable, __file__, "--worker", "--case", args.case,
-             "--seconds", str(args.seconds)],
+             "--seconds", str(args.seconds), "--settle", str(args.settle)],
             env=env, stdin=subprocess.PIPE, stdout=subprocess.PIPE, text=True))
 
     arb_stop = None

@@ -244,3 +244,23 @@ class TestConfigFlag:
             "nodeconfig": [{"name": "n1", "computepartition": "CPX"}]}))
         cfg = PluginConfig(node_name="n1", config_file=str(cfgfile))
         assert apply_node_config(cfg).compute_partition == "CPX"
+
+
+class TestNPSPartitionedMemory:
+    def test_distinct_banks_not_divided(self, tmp_path):
+        """NPS4/NPS8: partitions report their OWN memory banks (different
+        sizes in this fixture) — the shared-view division must NOT fire."""
+        import shutil
+
+        topo, pci, drm = make_cpx_tree(tmp_path)
+        # rewrite each partition's bank to a distinct size
+        for k in range(8):
+            nd = tmp_path / "topology" / "nodes" / str(k + 1)
+            (nd / "mem_banks" / "0" / "properties").write_text(
+                f"heap_type 1\nsize_in_bytes {(36 + k) * GB}\n")
+        gpus = enumerate_gpus(str(topo), str(pci), str(drm))
+        assert len(gpus) == 8
+        # UUIDs still disambiguated, memory untouched
+        assert len({g.uuid for g in gpus}) == 8
+        assert sorted(g.mem_bytes for g in gpus) == \
+            [(36 + k) * GB for k in range(8)]
