@@ -55,7 +55,6 @@ class FeedbackLoop:
             rk = e.region.get_recent_kernel()
             active = rk > ACTIVITY_THRESHOLD and bool(snap.procs)
             activity[e.key] = active
-            self._activity = activity
             # decay: an idle container's recent_kernel drifts to 0
             if rk > 0:
                 e.region.set_recent_kernel(rk - 1)
@@ -63,6 +62,7 @@ class FeedbackLoop:
                 if not uuid:
                     continue
                 by_device.setdefault(uuid, {}).setdefault(snap.priority, []).append(e)
+        self._activity = activity
 
         blocked: Set[str] = set()
         contended: Set[str] = set()

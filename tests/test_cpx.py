@@ -264,3 +264,23 @@ class TestNPSPartitionedMemory:
         assert len({g.uuid for g in gpus}) == 8
         assert sorted(g.mem_bytes for g in gpus) == \
             [(36 + k) * GB for k in range(8)]
+
+
+class TestPartitionSwitchErrors:
+    def test_unwritable_path_fails_cleanly(self, tmp_path):
+        """gpurun-style containers mount /sys read-only: the write must
+        fail cleanly (logged, False) and never raise.  Simulated with a
+        directory at the file path (root ignores mode bits, so chmod
+        cannot model EROFS)."""
+        card = tmp_path / "card0" / "device"
+        card.mkdir(parents=True)
+        f = card / "current_compute_partition"
+        f.mkdir()  # open(.., "w") -> IsADirectoryError (an OSError)
+        assert partition.write_mode(str(f), "CPX") is False
+
+    def test_write_already_in_mode_is_noop(self, tmp_path):
+        card = tmp_path / "card0" / "device"
+        card.mkdir(parents=True)
+        f = card / "current_compute_partition"
+        f.write_text("CPX")
+        assert partition.write_mode(str(f), "CPX")
