@@ -230,3 +230,87 @@ def test_oversubscription_end_to_end(tmp_path):
     assert lines[1]["err"] == 0                 # 300 GiB > physical succeeds
     assert lines[2]["managed"] >= 1             # went through hipMallocManaged
     plugin.stop()
+
+
+def test_cpx_partition_lifecycle_end_to_end(tmp_path):
+    """CPX node end to end: 8 XCD partitions register as typed devices,
+    the scheduler places a whole-partition pod (type-pinned via
+    use-gputype), and Allocate hands out ONLY that partition's render
+    node — hard isolation flowing through every protocol layer."""
+    from test_cpx import make_cpx_tree
+
+    from k8s_device_plugin_amd.device.amd import GPU_IN_USE_ANNO
+
+    client = FakeKubeClient()
+    client.add_node(NodeInfo(name="n1"))
+
+    topo, pci, drm = make_cpx_tree(tmp_path)
+    gpus = enumerate_gpus(str(topo), str(pci), str(drm))
+    rm = ResourceManager(gpus, split_count=2)
+    cfg = PluginConfig(
+        node_name="n1",
+        hook_path=str(tmp_path / "hook"),
+        plugin_socket_dir=str(tmp_path),
+        kubelet_socket=str(tmp_path / "kubelet.sock"),
+    )
+    plugin = VGPUDevicePlugin(cfg, rm, client)
+    plugin.serve()
+    register_once(client, "n1", rm)
+
+    sched = Scheduler(client)
+    sched.register_from_node_annotations_once()
+    # all 8 partitions visible to the scheduler with the CPX type
+    info = sched.node_manager.get_node("n1")
+    assert info is not None and len(info.devices) == 8
+    assert all(d.type.endswith("-CPX") for d in info.devices)
+
+    pod_obj = {
+        "kind": "Pod",
+        "metadata": {"name": "cpx-1", "namespace": "default",
+                     "uid": "uid-cpx-1",
+                     "annotations": {
+                         GPU_IN_USE_ANNO: "AMD-Instinct-MI355X-CPX"}},
+        "spec": {"containers": [{
+            "name": "main",
+            "resources": {"limits": {
+                "amd.com/gpu": "1",
+                "amd.com/gpucores": "100",   # whole partition
+            }},
+        }]},
+    }
+    pod = PodInfo.from_k8s(pod_obj)
+    client.add_pod(pod)
+    fr = sched.filter(pod, ["n1"])
+    assert fr.node_names == ["n1"], fr
+    br = sched.bind("cpx-1", "default", "n1")
+    assert not br.error
+
+    # which partition did the scheduler pick?
+    from k8s_device_plugin_amd.utils.codec import decode_container_devices
+    from k8s_device_plugin_amd.utils.types import IN_REQUEST_DEVICES
+
+    anno = client.get_pod("cpx-1").annotations[IN_REQUEST_DEVICES["AMD"]]
+    assigned = decode_container_devices(anno.split(";")[0])
+    part_uuid = assigned[0].uuid
+    part = rm.by_uuid(part_uuid)
+    assert part is not None and part.partition_count == 8
+
+    with grpc.insecure_channel(f"unix://{plugin.socket_path}") as ch:
+        stub = dp.DevicePluginClient(ch)
+        req = dp.AllocateRequest()
+        req.container_requests.add(devicesIDs=[f"{part_uuid}-0"])
+        resp = stub.Allocate(req)
+    ctr = resp.container_responses[0]
+    devpaths = {d.container_path for d in ctr.devices}
+    want = f"/dev/dri/renderD{128 + part.partition_index}"
+    assert want in devpaths
+    for k in range(8):
+        if k != part.partition_index:
+            assert f"/dev/dri/renderD{128 + k}" not in devpaths
+    envs = dict(ctr.envs)
+    assert envs["ROCR_VISIBLE_DEVICES"] == part_uuid
+    # whole partition: the hardware fences it, no mask env needed
+    assert "HSA_CU_MASK" not in envs
+    assert client.get_pod("cpx-1").annotations[BIND_PHASE_ANNO] == \
+        BIND_PHASE_SUCCESS
+    plugin.stop()
