@@ -1,17 +1,20 @@
 /* CU-percent soft throttle: token bucket on kernel launches, driven by a
  * utilization-feedback watcher thread.
  *
- * MI355X enforcement is two-layered (SURVEY.md §7 hard part 2):
- *  - HARD floor: the device plugin sets HSA_CU_MASK at Allocate so ROCr
- *    creates every queue on a fixed CU subset of the 256-CU chip (the DCU
- *    vdev cu_mask design, reference pkg/device-plugin/hygon/dcu/corealloc.go)
- *    — no interceptor work needed, enforced by the runtime.
- *  - SOFT ceiling (this file): a shared token bucket in the region paced to
- *    the container's CU percent.  Launch credits are workgroup counts; a
- *    watcher thread measures GPU utilization and adjusts the fill rate so
- *    the container converges on its share even when the CU mask is disabled
- *    or work is launch-bound.  Reference analog: rate_limiter +
- *    utilization_watcher in the CUDA hook (SURVEY.md §2.6 "Core throttle").
+ * MI355X enforcement is layered (SURVEY.md §7 hard part 2):
+ *  - HARD partition: CPX compute partitioning (plugin/partition.py) — each
+ *    XCD is its own device and Allocate mounts only that partition's
+ *    render node; nothing for the interceptor to do.  (HSA_CU_MASK is
+ *    still injected as a best-effort layer, but KFD ignores per-queue
+ *    masks on multi-XCD gfx9 — tests/test_gpu.py documents this.)
+ *  - SOFT ceiling (this file): a shared token bucket in the region paced
+ *    to the container's CU percent.  Launch credits are WAVEFRONTS
+ *    (grid x ceil(block/64), the CDNA4 issue unit, so fat-workgroup
+ *    kernels pay proportionally); the fill rate is set by the monitor's
+ *    arbitrated scale when fresh (monitor/arbiter.py — token-bound
+ *    max-min fairness across co-located pods) and by a local
+ *    utilization-feedback loop otherwise.  Reference analog:
+ *    rate_limiter + utilization_watcher (SURVEY.md §2.6 "Core throttle").
  *
  * Tunables (env):
  *  VGPU_TOKEN_RATE   fixed tokens/sec, disables feedback (deterministic tests)
@@ -36,10 +39,12 @@ typedef int hipError_t;
 
 #define NSEC 1000000000ULL
 #define REFILL_INTERVAL_NS (50ULL * 1000 * 1000) /* 50 ms */
-/* A full MI355X retires small workgroups at O(10^6)/s when launch-bound;
- * credits are deliberately generous — the throttle should bind utilization,
- * not launch count, and the feedback loop re-scales it anyway. */
-#define RATE_FULL 4000000.0 /* workgroups/sec at 100% */
+/* Nominal full-chip issue rate in wavefronts/sec.  This is a CALIBRATION
+ * CONSTANT, not a measurement: real workloads sit orders of magnitude off
+ * it, and both control loops (the monitor's arbiter and the local EMA)
+ * re-scale it — the slow-start dynamics in monitor/arbiter.py exist
+ * precisely to traverse that miscalibration quickly. */
+#define RATE_FULL 4000000.0 /* wavefronts/sec at 100% */
 #define BUCKET_SECONDS 0.25 /* cap: a quarter second of fill */
 
 static pthread_t g_watcher;
