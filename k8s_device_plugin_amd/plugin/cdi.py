@@ -58,6 +58,18 @@ def generate_spec(gpus: List[PhysicalGPU], hook_path: str = "") -> Dict:
                 },
             ]
         devices.append({"name": g.uuid, "containerEdits": edits})
+    if gpus:
+        # composite "all" device (the reference's nvcdi generates one,
+        # cdi/cdi.go): every GPU's nodes behind a single CDI name
+        seen = set()
+        union = []
+        for g in gpus:
+            for node in _device_nodes(g):
+                if node["path"] not in seen:
+                    seen.add(node["path"])
+                    union.append(node)
+        devices.append({"name": "all",
+                        "containerEdits": {"deviceNodes": union}})
     return {
         "cdiVersion": CDI_VERSION,
         "kind": CDI_KIND,

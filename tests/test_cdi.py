@@ -19,7 +19,7 @@ def test_spec_shape():
     spec = cdi.generate_spec([mkgpu(0), mkgpu(1)], hook_path="/usr/local")
     assert spec["kind"] == "amd.com/gpu"
     assert spec["cdiVersion"] == "0.5.0"
-    assert len(spec["devices"]) == 2
+    assert len(spec["devices"]) == 3  # 2 GPUs + the composite "all"
     d0 = spec["devices"][0]
     assert d0["name"] == "GPU-0000000000000000"
     paths = [n["path"] for n in d0["containerEdits"]["deviceNodes"]]
@@ -42,7 +42,7 @@ def test_write_spec_atomic(tmp_path):
     # rewrite over existing
     cdi.write_spec([mkgpu(0), mkgpu(1)], spec_dir=str(tmp_path))
     spec = json.loads(open(path).read())
-    assert len(spec["devices"]) == 2
+    assert len(spec["devices"]) == 3  # 2 GPUs + the composite "all"
 
 
 def test_annotation_roundtrip():
@@ -51,3 +51,21 @@ def test_annotation_roundtrip():
     assert cdi.parse_annotation(ann["cdi.k8s.io/vgpu-amd"]) == ["GPU-a", "GPU-b"]
     assert cdi.annotations([]) == {}
     assert cdi.parse_annotation("nvidia.com/gpu=X") == []
+
+
+def test_all_composite_device(tmp_path):
+    """nvcdi-style composite: one CDI device named "all" holding every
+    GPU's nodes exactly once (/dev/kfd deduplicated)."""
+    from test_plugin import make_kfd_tree
+
+    from k8s_device_plugin_amd.plugin import cdi
+    from k8s_device_plugin_amd.plugin.kfd import enumerate_gpus
+
+    topo, pci = make_kfd_tree(tmp_path, n_gpus=2)
+    gpus = enumerate_gpus(str(topo), str(pci))
+    spec = cdi.generate_spec(gpus)
+    byname = {d["name"]: d for d in spec["devices"]}
+    assert "all" in byname
+    paths = [n["path"] for n in byname["all"]["containerEdits"]["deviceNodes"]]
+    assert paths.count("/dev/kfd") == 1
+    assert "/dev/dri/renderD128" in paths and "/dev/dri/renderD129" in paths
