@@ -15,7 +15,7 @@ from __future__ import annotations
 
 import logging
 import time
-from typing import Callable, Dict, List, Optional, Set, Tuple
+from typing import Callable, Dict, List, Optional, Set
 
 from .arbiter import ScaleArbiter
 from .pathmon import ContainerEntry, PathMonitor

@@ -18,7 +18,7 @@ import os
 import struct
 from dataclasses import dataclass
 from pathlib import Path
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 _DEFAULT_LIB = Path(__file__).resolve().parent.parent / "csrc" / "libvgpu-hip.so"
 

@@ -9,7 +9,6 @@ Responds with a JSONPatch AdmissionReview.
 from __future__ import annotations
 
 import base64
-import copy
 import json
 import logging
 from typing import Optional

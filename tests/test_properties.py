@@ -137,3 +137,27 @@ def test_decoders_never_crash_on_garbage(s):
     except CodecError:
         pass
     decode_node_xgmi(s)  # xgmi decode is total: any text -> dict
+
+
+def test_codec_roundtrip_with_cpx_partition_uuids():
+    """CPX partition UUIDs carry a '.' suffix (GPU-<id>.<k>) — the wire
+    codec's separators are ',' ':' ';' so dotted IDs must round-trip
+    through both the node and container encodings."""
+    from k8s_device_plugin_amd.utils.codec import (
+        decode_container_devices,
+        decode_node_devices,
+        encode_container_devices,
+        encode_node_devices,
+    )
+    from k8s_device_plugin_amd.utils.types import ContainerDevice, DeviceInfo
+
+    devs = [DeviceInfo(id=f"GPU-00abc.{k}", count=2, devmem=36864,
+                       devcore=100, type="AMD-Instinct-MI355X-CPX",
+                       numa=0, health=True, index=k) for k in range(8)]
+    back = decode_node_devices(encode_node_devices(devs))
+    assert [d.id for d in back] == [d.id for d in devs]
+
+    cds = [ContainerDevice(uuid=f"GPU-00abc.{k}", type="AMD",
+                           usedmem=36864, usedcores=100) for k in range(3)]
+    back2 = decode_container_devices(encode_container_devices(cds))
+    assert [c.uuid for c in back2] == [c.uuid for c in cds]
